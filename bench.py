@@ -1,0 +1,139 @@
+"""Benchmark entry point (driver contract).
+
+Measures the flagship DSIN training step — full model (AE + probclass +
+siFinder NCC search + siNet fusion), bf16 compute, 320x960 synthetic
+KITTI-shaped stereo pairs, batch 1 per GPU (the reference's SI training
+batch, src/AE.py:26) — and prints ONE JSON line with the whole-job
+imgs/sec aggregate. Metric and config follow BASELINE.json
+("imgs/sec train @320x960 per GPU", synthetic data, random-init weights).
+
+Launch: `python bench.py --gpus N --steps K --warmup W`; for N>1 the driver
+uses torch.distributed.run with one rank per GPU over RCCL; we read
+RANK/LOCAL_RANK/WORLD_SIZE from the env. Weak scaling: per-GPU work fixed.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from dsin_amd import config as config_mod
+from dsin_amd.data import SyntheticStereo
+from dsin_amd.models import DSIN
+from dsin_amd.parallel import init_distributed, is_distributed, rank, world_size
+from dsin_amd.training import Trainer
+
+BASELINE_VALUE = None  # reference publishes no throughput number
+
+
+def build_configs(args):
+    cur = os.path.dirname(os.path.abspath(__file__))
+    ae, _ = config_mod.parse(os.path.join(cur, "run_configs", "ae_run_configs"))
+    pc, _ = config_mod.parse(os.path.join(cur, "run_configs", "pc_run_configs"))
+    ae.crop_size = (args.height, args.width)
+    ae.AE_only = bool(args.ae_only)
+    ae.load_model = False
+    ae.train_model = True
+    ae.test_model = False
+    return ae, pc
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--height", type=int, default=320)
+    ap.add_argument("--width", type=int, default=960)
+    ap.add_argument("--batch", type=int, default=1)
+    ap.add_argument("--ae-only", action="store_true",
+                    help="benchmark config 2 (AE without side information)")
+    ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--device", type=str, default=None)
+    args = ap.parse_args()
+
+    local_rank = init_distributed()
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device("cpu")
+    n_ranks = world_size()
+
+    ae_config, pc_config = build_configs(args)
+    torch.manual_seed(1234 + rank())
+    model = DSIN(ae_config, pc_config).to(device)
+    trainer = Trainer(model, ae_config, pc_config, num_training_imgs=1576,
+                     device=device, autocast_bf16=(args.dtype == "bf16"),
+                     ddp_comm_dtype=None)
+
+    batch = args.batch if args.ae_only else 1
+    gen = SyntheticStereo(args.height, args.width, batch_size=batch,
+                          seed=1234 + rank(), device=str(device))
+    # pre-generate batches so host-side synthesis stays out of the timed region
+    n_batches = min(args.warmup + args.steps, 16)
+    batches = [gen.next_batch() for _ in range(n_batches)]
+
+    def step(i):
+        x, y = batches[i % n_batches]
+        trainer.train_step(x, y if not args.ae_only else None)
+
+    for i in range(args.warmup):
+        step(i)
+
+    def barrier_sync():
+        if is_distributed():
+            torch.distributed.barrier()
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if is_distributed():
+        t = torch.tensor([elapsed], device=device if device.type == "cuda" else None)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    imgs = args.steps * batch * n_ranks
+    value = imgs / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank() == 0:
+        result = {
+            "metric": "imgs/sec train @320x960 per GPU; MS-SSIM @0.02 bpp, 1/2/4/8 MI355X",
+            "value": value,
+            "unit": "imgs/sec",
+            "n_gpus": n_ranks,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": BASELINE_VALUE,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "DSIN (AE+probclass+siFinder+siNet)" if not args.ae_only
+                         else "DSIN AE-only",
+                "global_batch": batch * n_ranks,
+                "crop": [args.height, args.width],
+                "seq_len": None,
+                "parallelism": f"dp{n_ranks}",
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

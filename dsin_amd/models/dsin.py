@@ -1,0 +1,200 @@
+"""DSIN model orchestrator — the reference's `AE` class rebuilt as an eager
+PyTorch module (/root/reference/src/AE.py:10-250).
+
+Owns encoder, decoder, probclass (entropy model), SI search and siNet fusion,
+and assembles the combined rate-distortion + side-information loss:
+
+  loss_train = (1-si_w)*d_loss + beta*max(H_soft - H_target, 0) + reg
+               + si_w * L1(x, x_with_si)                      (src/AE.py:80-99)
+
+Gradient topology kept from the reference:
+  * PC input is stop_gradient(qbar) (src/AE.py:74) — rate loss reaches the
+    encoder only through the heatmap in H_mask;
+  * y_syn enters siNet under stop_gradient (src/AE.py:67); the SI search
+    itself is non-trainable (src/siFinder.py:3-5);
+  * the straight-through estimator lives inside the quantize op.
+
+The train step additionally runs a no-grad eval-mode forward of the
+autoencoder on y to produce y_dec (src/AE.py:110,150-152) — the SI search
+correlates *decoded* x against *decoded* y.
+"""
+
+from __future__ import annotations
+
+from contextlib import nullcontext
+from typing import Dict, NamedTuple, Optional
+
+import torch
+import torch.nn as nn
+
+from ..losses.distortions import Distortions, bitcost_to_bpp, get_loss
+from .autoencoder import Decoder, Encoder
+from .probclass import ProbClass
+from .sifinder import SiFinder
+from .sinet import SiNet
+from .. import ops
+
+
+class StepOutput(NamedTuple):
+    loss: torch.Tensor
+    bpp: torch.Tensor
+    x_dec: torch.Tensor
+    x_with_si: Optional[torch.Tensor]
+
+
+class DSIN(nn.Module):
+    def __init__(self, ae_config, pc_config):
+        super().__init__()
+        self.ae_config = ae_config
+        self.pc_config = pc_config
+        self.ae_only = bool(ae_config.AE_only)
+        self.si_weight = 0.0 if self.ae_only else float(ae_config.si_weight)
+
+        self.encoder = Encoder(ae_config)
+        self.decoder = Decoder(ae_config)
+        self.probclass = ProbClass(pc_config, num_centers=ae_config.num_centers)
+        if not self.ae_only:
+            self.sifinder = SiFinder(ae_config)
+            self.sinet = SiNet()
+        else:
+            self.sifinder = None
+            self.sinet = None
+
+        self.reg_factor = float(ae_config.regularization_factor)
+        self.use_centers_for_padding = bool(pc_config.use_centers_for_padding)
+
+    # -- forward pieces ----------------------------------------------------
+
+    def autoencode(self, x: torch.Tensor):
+        z = self.encoder(x)
+        x_dec = self.decoder(z.qbar)
+        return z, x_dec
+
+    @torch.no_grad()
+    def create_y_dec(self, y: torch.Tensor) -> torch.Tensor:
+        """Eval-mode (EMA batch-norm) autoencoder pass over the side image
+        (reference src/AE.py:150-152 runs with is_training=False)."""
+        was_training = self.training
+        self.eval()
+        try:
+            z = self.encoder(y)
+            y_dec = self.decoder(z.qbar)
+        finally:
+            self.train(was_training)
+        return y_dec
+
+    def _pad_value(self) -> torch.Tensor:
+        if self.use_centers_for_padding:
+            return self.encoder.quantizer.centers[0].detach()
+        return torch.zeros((), device=self.encoder.quantizer.centers.device)
+
+    def regularization_loss(self) -> torch.Tensor:
+        """L2 regularizers: factor * sum(w^2)/2 over encoder+decoder conv
+        weights, the centers term, plus PC when enabled (reference
+        src/Distortions_imgcomp.py:129-136, src/quantizer_imgcomp.py:18-24)."""
+        dev = self.encoder.quantizer.centers.device
+        s = torch.zeros((), device=dev)
+        if self.reg_factor:
+            for mod in (self.encoder, self.decoder):
+                for m in mod.modules():
+                    if isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
+                        s = s + 0.5 * (m.weight.float() ** 2).sum()
+            s = s * self.reg_factor
+        s = s + self.encoder.quantizer.regularization_loss()
+        s = s + self.probclass.regularization_loss()
+        return s
+
+    def side_information(self, x_dec: torch.Tensor, y: torch.Tensor,
+                         y_dec: torch.Tensor) -> torch.Tensor:
+        """x_with_si from decoded x + original/decoded y (src/AE.py:58-69)."""
+        with torch.no_grad():
+            y_syn = self.sifinder(x_dec.detach().float(), y.float(), y_dec.float())
+        cat = torch.cat([ops.kitti_normalize(x_dec),
+                         ops.kitti_normalize(y_syn.to(x_dec.dtype)).detach()], dim=1)
+        return ops.kitti_denormalize(self.sinet(cat))
+
+    # -- losses ------------------------------------------------------------
+
+    def compute_losses(self, x: torch.Tensor, y: Optional[torch.Tensor],
+                       y_dec: Optional[torch.Tensor], detach_pc_input: bool = True):
+        """One full forward + loss assembly. Returns dict of scalars/tensors."""
+        z, x_dec = self.autoencode(x)
+
+        x32, xd32 = x.float(), x_dec.float()
+        d = Distortions(self.ae_config, x32, xd32, is_training=True)
+
+        pc_in = z.qbar.detach() if detach_pc_input else z.qbar
+        bc = self.probclass.bitcost(pc_in.float(), z.symbols, self._pad_value())
+        bpp = bitcost_to_bpp(bc, x)
+
+        heatmap = z.heatmap.float() if z.heatmap is not None else None
+        reg = self.regularization_loss()
+        total, H_real, pc_loss = get_loss(
+            self.ae_config, (1.0 - self.si_weight) * d.d_loss_scaled, bc, heatmap, reg)
+
+        x_with_si = None
+        loss_sinet = torch.zeros((), device=x.device)
+        if not self.ae_only:
+            assert y is not None and y_dec is not None
+            x_with_si = self.side_information(x_dec, y, y_dec)
+            loss_sinet = (x32 - x_with_si.float()).abs().mean()
+
+        loss = total + self.si_weight * loss_sinet
+        return {
+            "loss": loss, "bpp": bpp, "x_dec": x_dec, "x_with_si": x_with_si,
+            "H_real": H_real, "pc_loss": pc_loss, "d_loss": d.d_loss_scaled,
+            "reg": reg, "loss_sinet": loss_sinet, "symbols": z.symbols,
+        }
+
+    # -- reference step API (src/AE.py:108-152) ----------------------------
+
+    def train_losses(self, x: torch.Tensor, y: Optional[torch.Tensor]):
+        """Forward for one training step (autograd on). Caller does
+        backward + optimizer steps (see training/trainer.py)."""
+        self.train()
+        y_dec = None
+        if not self.ae_only:
+            y_dec = self.create_y_dec(y)
+        return self.compute_losses(x, y, y_dec, detach_pc_input=True)
+
+    @torch.no_grad()
+    def validate_loss(self, x: torch.Tensor, y: Optional[torch.Tensor]) -> torch.Tensor:
+        self.eval()
+        y_dec = self.create_y_dec(y) if not self.ae_only else None
+        out = self.compute_losses(x, y, y_dec, detach_pc_input=False)
+        return out["loss"]
+
+    @torch.no_grad()
+    def reconstruct(self, x: torch.Tensor, y: Optional[torch.Tensor]):
+        """Test-time reconstruction (reference siNet_get_reconstructed,
+        src/AE.py:132-148): returns (y_dec, y_syn, x_dec, x_with_si, bpp)."""
+        self.eval()
+        z, x_dec = self.autoencode(x)
+        bc = self.probclass.bitcost(z.qbar.float(), z.symbols, self._pad_value())
+        bpp = bitcost_to_bpp(bc, x)
+        if self.ae_only:
+            return None, None, x_dec, torch.zeros_like(x_dec), bpp
+        y_dec = self.create_y_dec(y)
+        y_syn = self.sifinder(x_dec.float(), y.float(), y_dec.float())
+        cat = torch.cat([ops.kitti_normalize(x_dec),
+                         ops.kitti_normalize(y_syn.to(x_dec.dtype))], dim=1)
+        x_with_si = ops.kitti_denormalize(self.sinet(cat))
+        return y_dec, y_syn, x_dec, x_with_si, bpp
+
+    # -- checkpoint scope groups (reference src/AE.py:154-175) --------------
+
+    def state_groups(self) -> Dict[str, nn.Module]:
+        groups = {"encoder": self.encoder, "decoder": self.decoder,
+                  "imgcomp": self.probclass}
+        if self.sinet is not None:
+            groups["siNetwork"] = self.sinet
+        return groups
+
+    def param_groups(self):
+        """(ae_params, pc_params) for the two optimizers: PC variables get
+        their own Adam; everything else (enc, dec, centers, siNet) the default
+        one (reference src/AE.py:177-191)."""
+        pc_params = list(self.probclass.parameters())
+        pc_ids = {id(p) for p in pc_params}
+        ae_params = [p for p in self.parameters() if id(p) not in pc_ids]
+        return ae_params, pc_params

@@ -1,0 +1,94 @@
+"""End-to-end training integration tests on CPU (BASELINE config 1:
+AE encoder+quantizer+decoder forward on 64x64 random crops, CPU eager)."""
+
+import torch
+import pytest
+
+from dsin_amd.data import SyntheticStereo
+from dsin_amd.models import DSIN
+from dsin_amd.training import Trainer
+from dsin_amd.training.helpers import lr_at_step, num_itr_per_epoch
+
+
+def test_baseline_config1_forward(small_ae_config, pc_config):
+    cfg = small_ae_config.clone(crop_size=(64, 64), y_patch_size=(16, 16))
+    torch.manual_seed(0)
+    m = DSIN(cfg, pc_config)
+    x = torch.rand(1, 3, 64, 64) * 255
+    z, x_dec = m.autoencode(x)
+    assert x_dec.shape == x.shape
+    assert z.qbar.shape == (1, 32, 8, 8)
+
+
+def test_two_train_steps_finite(small_ae_config, pc_config):
+    torch.manual_seed(0)
+    m = DSIN(small_ae_config, pc_config)
+    tr = Trainer(m, small_ae_config, pc_config, num_training_imgs=1576)
+    gen = SyntheticStereo(64, 96)
+    losses = []
+    for _ in range(2):
+        x, y = gen.next_batch()
+        loss, bpp = tr.train_step(x, y)
+        assert torch.isfinite(loss) and torch.isfinite(bpp)
+        losses.append(float(loss))
+    assert tr.global_step == 2
+
+
+def test_ae_only_mode(small_ae_config, pc_config):
+    cfg = small_ae_config.clone(AE_only=True, batch_size=2)
+    torch.manual_seed(0)
+    m = DSIN(cfg, pc_config)
+    assert m.sinet is None and m.si_weight == 0.0
+    tr = Trainer(m, cfg, pc_config, num_training_imgs=1576)
+    x = torch.rand(2, 3, 64, 96) * 255
+    loss, bpp = tr.train_step(x, None)
+    assert torch.isfinite(loss)
+
+
+def test_validate_and_reconstruct(small_ae_config, pc_config):
+    torch.manual_seed(0)
+    m = DSIN(small_ae_config, pc_config)
+    gen = SyntheticStereo(64, 96)
+    x, y = gen.next_batch()
+    v = m.validate_loss(x, y)
+    assert torch.isfinite(v)
+    y_dec, y_syn, x_dec, x_with_si, bpp = m.reconstruct(x, y)
+    assert x_with_si.shape == x.shape and torch.isfinite(bpp)
+
+
+def test_lr_schedule_staircase(small_ae_config):
+    itr_ep = num_itr_per_epoch(1, 1, 1576, False)
+    assert itr_ep == 1576
+    decay_steps = itr_ep * 20
+    assert lr_at_step(small_ae_config, 0, itr_ep) == pytest.approx(1e-4)
+    assert lr_at_step(small_ae_config, decay_steps - 1, itr_ep) == pytest.approx(1e-4)
+    assert lr_at_step(small_ae_config, decay_steps, itr_ep) == pytest.approx(1e-5)
+    assert lr_at_step(small_ae_config, 2 * decay_steps, itr_ep) == pytest.approx(1e-6)
+
+
+def test_imagenet_epoch_for_ae_only():
+    assert num_itr_per_epoch(1, 1, 1576, True) == 1_281_000
+
+
+def test_two_optimizer_split(small_ae_config, pc_config):
+    m = DSIN(small_ae_config, pc_config)
+    ae_params, pc_params = m.param_groups()
+    n_pc = sum(p.numel() for p in pc_params)
+    assert 20e3 < n_pc < 28e3  # probclass only
+    ids = {id(p) for p in ae_params} | {id(p) for p in pc_params}
+    assert len(ids) == len(ae_params) + len(pc_params)  # disjoint
+    assert ids == {id(p) for p in m.parameters()}       # complete
+
+
+def test_gradient_isolation_sinet_vs_search(small_ae_config, pc_config):
+    """siNet loss must backprop into the DECODER through x_dec, but not into
+    anything through y_syn (stop_gradient, reference src/AE.py:67)."""
+    torch.manual_seed(0)
+    m = DSIN(small_ae_config, pc_config)
+    gen = SyntheticStereo(64, 96)
+    x, y = gen.next_batch()
+    out = m.train_losses(x, y)
+    loss_sinet = out["loss_sinet"]
+    loss_sinet.backward()
+    dec_grad = m.decoder.from_bn.conv.weight.grad
+    assert dec_grad is not None and dec_grad.abs().sum() > 0
