@@ -1,0 +1,28 @@
+// Python bindings for the dsin_amd HIP/CDNA4 kernels.
+#include <torch/extension.h>
+
+namespace dsin {
+std::tuple<torch::Tensor, torch::Tensor> quantize_fwd(torch::Tensor x,
+                                                      torch::Tensor centers,
+                                                      double sigma);
+std::tuple<torch::Tensor, torch::Tensor> quantize_bwd(torch::Tensor g,
+                                                      torch::Tensor x,
+                                                      torch::Tensor centers,
+                                                      double sigma);
+torch::Tensor bitcost_ce_fwd(torch::Tensor logits, torch::Tensor symbols);
+torch::Tensor bitcost_ce_bwd(torch::Tensor g, torch::Tensor logits,
+                             torch::Tensor symbols);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ncc_search(
+    torch::Tensor x_dec, torch::Tensor y_dec, torch::Tensor y_orig, int64_t ph,
+    int64_t pw, bool use_mask);
+torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B);
+}  // namespace dsin
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("quantize_fwd", &dsin::quantize_fwd, "fused soft quantizer forward");
+  m.def("quantize_bwd", &dsin::quantize_bwd, "fused soft quantizer backward");
+  m.def("bitcost_ce_fwd", &dsin::bitcost_ce_fwd, "bitcost cross-entropy fwd");
+  m.def("bitcost_ce_bwd", &dsin::bitcost_ce_bwd, "bitcost cross-entropy bwd");
+  m.def("ncc_search", &dsin::ncc_search, "streaming NCC side-info search");
+  m.def("mfma_selftest", &dsin::mfma_selftest, "MFMA 16x16x32 layout check");
+}

@@ -190,7 +190,7 @@ def gaussian_mask_value(num_patches_w: int, ph: int, pw: int, H: int, W: int,
 
 def ncc_search_ref(x_dec: torch.Tensor, y_dec: torch.Tensor, y_orig: torch.Tensor,
                    ph: int, pw: int, use_mask: bool = True,
-                   eps: float = 0.0) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+                   eps: float = 1e-10) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Full SI search for ONE image (the reference loops images at batch 1,
     src/siFull_img.py:15-39). All inputs (3, H, W) [y_* may be larger than
     x_dec in general; equal here]. Returns (y_syn (3,Hx,Wx), rows (P,), cols (P,)).
