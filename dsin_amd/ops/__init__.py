@@ -85,10 +85,13 @@ class _QuantizeFn(torch.autograd.Function):
 
 def quantize(x: torch.Tensor, centers: torch.Tensor, sigma: float = 1.0
              ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Returns (qbar, symbols). See reference src/quantizer_imgcomp.py:37-100."""
+    """Returns (qbar, symbols). See reference src/quantizer_imgcomp.py:37-100.
+    Always computed in fp32 (the bottleneck is tiny; bf16 autocast inputs are
+    upcast here and gradients flow back through the cast)."""
     if x.is_cuda:
-        return _QuantizeFn.apply(x.contiguous(), centers.contiguous(), sigma)
-    qbar, _, _, symbols = ref.quantize_ref(x, centers, sigma)
+        return _QuantizeFn.apply(x.float().contiguous(),
+                                 centers.float().contiguous(), sigma)
+    qbar, _, _, symbols = ref.quantize_ref(x.float(), centers, sigma)
     return qbar, symbols
 
 
@@ -125,10 +128,12 @@ class _BitcostCEFn(torch.autograd.Function):
 
 
 def bitcost_ce(logits: torch.Tensor, symbols: torch.Tensor) -> torch.Tensor:
-    """logits (N, L, C, H, W), symbols (N, C, H, W) -> bits (N, C, H, W)."""
+    """logits (N, L, C, H, W), symbols (N, C, H, W) -> bits (N, C, H, W).
+    fp32 compute (autocast bf16 logits are upcast here)."""
     if logits.is_cuda:
-        return _BitcostCEFn.apply(logits.contiguous(), symbols.contiguous())
-    return ref.bitcost_ce_ref(logits, symbols)
+        return _BitcostCEFn.apply(logits.float().contiguous(),
+                                  symbols.contiguous())
+    return ref.bitcost_ce_ref(logits.float(), symbols)
 
 
 # ---------------------------------------------------------------------------
