@@ -54,6 +54,8 @@ def main():
                     help="benchmark config 2 (AE without side information)")
     ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph capture of the train step")
     args = ap.parse_args()
 
     local_rank = init_distributed()
@@ -70,6 +72,7 @@ def main():
     model = DSIN(ae_config, pc_config).to(device)
     trainer = Trainer(model, ae_config, pc_config, num_training_imgs=1576,
                      device=device, autocast_bf16=(args.dtype == "bf16"),
+                     use_cuda_graph=(not args.no_graph and device.type == "cuda"),
                      ddp_comm_dtype=None)
 
     batch = args.batch if args.ae_only else 1

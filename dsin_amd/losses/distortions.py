@@ -41,18 +41,20 @@ def psnr_per_image(x: torch.Tensor, y: torch.Tensor, cast_to_int: bool) -> torch
 
 
 class Distortions:
-    """Computed lazily on construction like the reference class (:7-34)."""
+    """Lazy metric evaluation: TF builds all metric nodes but only computes
+    the fetched ones (the reference's sess.run prunes the graph); eager
+    mirrors that by evaluating each metric on first attribute access.
+    `d_loss_scaled` (the minimized distortion) is computed eagerly."""
 
     def __init__(self, config, x: torch.Tensor, x_out: torch.Tensor, is_training: bool):
         minimize_for = config.distortion_to_minimize
         assert minimize_for in ("mae", "mse", "psnr", "ms_ssim")
-        int_psnr = (not is_training) or minimize_for != "psnr"
-        int_mse = (not is_training) or minimize_for != "mse"
-        int_mae = (not is_training) or minimize_for != "mae"
-        self.mae = mae_per_image(x, x_out, int_mae).mean()
-        self.mse = mse_per_image(x, x_out, int_mse).mean()
-        self.psnr = psnr_per_image(x, x_out, int_psnr).mean()
-        self.ms_ssim = multiscale_ssim(x, x_out) if minimize_for == "ms_ssim" else None
+        self._x, self._xo = x, x_out
+        self._int_psnr = (not is_training) or minimize_for != "psnr"
+        self._int_mse = (not is_training) or minimize_for != "mse"
+        self._int_mae = (not is_training) or minimize_for != "mae"
+        self._cache = {}
+        self._minimize_for = minimize_for
 
         if minimize_for == "mae":
             self.d_loss_scaled = self.mae
@@ -62,6 +64,30 @@ class Distortions:
             self.d_loss_scaled = config.K_psnr - self.psnr
         else:
             self.d_loss_scaled = config.K_ms_ssim * (1.0 - self.ms_ssim)
+
+    @property
+    def mae(self):
+        if "mae" not in self._cache:
+            self._cache["mae"] = mae_per_image(self._x, self._xo, self._int_mae).mean()
+        return self._cache["mae"]
+
+    @property
+    def mse(self):
+        if "mse" not in self._cache:
+            self._cache["mse"] = mse_per_image(self._x, self._xo, self._int_mse).mean()
+        return self._cache["mse"]
+
+    @property
+    def psnr(self):
+        if "psnr" not in self._cache:
+            self._cache["psnr"] = psnr_per_image(self._x, self._xo, self._int_psnr).mean()
+        return self._cache["psnr"]
+
+    @property
+    def ms_ssim(self):
+        if "ms_ssim" not in self._cache:
+            self._cache["ms_ssim"] = multiscale_ssim(self._x, self._xo)
+        return self._cache["ms_ssim"]
 
 
 def get_loss(config, d_loss_scaled: torch.Tensor, bc: torch.Tensor,

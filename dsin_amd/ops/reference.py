@@ -103,7 +103,11 @@ def pad_for_probclass_ref(q: torch.Tensor, pad: int, pad_value: torch.Tensor) ->
     """
     n, c, h, w = q.shape
     out = q.new_empty(n, c + pad, h + 2 * pad, w + 2 * pad)
-    out.fill_(float(pad_value.detach()) if isinstance(pad_value, torch.Tensor) else pad_value)
+    if isinstance(pad_value, torch.Tensor):
+        # device-side broadcast fill: no host sync (.item()) in the hot loop
+        out.copy_(pad_value.detach().to(out.dtype).expand_as(out))
+    else:
+        out.fill_(pad_value)
     out[:, pad:, pad:-pad, pad:-pad] = q
     return out
 
