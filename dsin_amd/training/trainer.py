@@ -34,8 +34,9 @@ def _make_optimizer(config, params, device, capturable: bool):
         if capturable:
             lr_t = torch.tensor(lr, device=device)
             return torch.optim.Adam(params, lr=lr_t, betas=(0.9, 0.999),
-                                    eps=1e-8, capturable=True)
-        return torch.optim.Adam(params, lr=lr, betas=(0.9, 0.999), eps=1e-8)
+                                    eps=1e-8, capturable=True, foreach=True)
+        return torch.optim.Adam(params, lr=lr, betas=(0.9, 0.999), eps=1e-8,
+                                foreach=True)
     return create_optimizer(config, params)
 
 
@@ -103,7 +104,9 @@ class Trainer:
 
     def _autocast(self):
         if self.autocast_bf16:
-            return torch.autocast(device_type="cuda", dtype=torch.bfloat16)
+            # autocast's weight-cast cache is incompatible with graph capture
+            return torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                                  cache_enabled=not self.use_cuda_graph)
         return contextlib.nullcontext()
 
     def _step_inner(self, x: torch.Tensor, y: Optional[torch.Tensor]):
@@ -135,7 +138,9 @@ class Trainer:
                 self._static_out = self._step_inner(self._static_x, self._static_y)
             self._graph = graph
         except Exception as e:  # pragma: no cover - device specific
-            warnings.warn(f"hipGraph capture failed, staying eager: {e}")
+            import traceback
+            warnings.warn("hipGraph capture failed, staying eager: "
+                          + "".join(traceback.format_exception(e)))
             self._graph_failed = True
             self._graph = None
 
