@@ -67,7 +67,8 @@ def heatmap3d_ref(bottleneck: torch.Tensor) -> torch.Tensor:
     """
     C = bottleneck.shape[1] - 1
     h2d = torch.sigmoid(bottleneck[:, 0]) * C               # (N,H,W)
-    c = torch.arange(C, dtype=bottleneck.dtype, device=bottleneck.device).view(1, C, 1, 1)
+    c = _cached_const(tuple(float(i) for i in range(C)), bottleneck.device,
+                      bottleneck.dtype, (1, C, 1, 1))
     return (h2d.unsqueeze(1) - c).clamp(0.0, 1.0)
 
 
@@ -75,10 +76,24 @@ def heatmap3d_ref(bottleneck: torch.Tensor) -> torch.Tensor:
 # Fixed KITTI normalization (reference src/AE.py:222-248)
 # ---------------------------------------------------------------------------
 
+_CONST_CACHE = {}
+
+
+def _cached_const(values, device, dtype, shape):
+    """Per-(values, device, dtype) cached constant tensor. Avoids a pageable
+    host->device copy per call — which is also forbidden inside hipGraph
+    capture (hipErrorStreamCaptureUnsupported)."""
+    key = (tuple(values), str(device), dtype, tuple(shape))
+    t = _CONST_CACHE.get(key)
+    if t is None:
+        t = torch.tensor(values, device=device, dtype=dtype).view(shape)
+        _CONST_CACHE[key] = t
+    return t
+
+
 def _mean_std(device, dtype):
-    mean = torch.tensor(KITTI_MEAN, device=device, dtype=dtype).view(1, 3, 1, 1)
-    std = torch.tensor(KITTI_STD, device=device, dtype=dtype).view(1, 3, 1, 1)
-    return mean, std
+    return (_cached_const(KITTI_MEAN, device, dtype, (1, 3, 1, 1)),
+            _cached_const(KITTI_STD, device, dtype, (1, 3, 1, 1)))
 
 
 def kitti_normalize(x: torch.Tensor) -> torch.Tensor:
@@ -159,11 +174,11 @@ def _sifinder_norm(x: torch.Tensor) -> torch.Tensor:
     """Per-channel fixed normalization used inside the SI search
     (reference src/siFinder.py:56-73: (v - mean) / std with the std values
     stored under the name `variances`)."""
-    mean = torch.tensor(KITTI_MEAN, device=x.device, dtype=x.dtype)
-    std = torch.tensor(KITTI_STD_SIFINDER, device=x.device, dtype=x.dtype)
     shape = [1] * x.dim()
     shape[-3] = 3
-    return (x - mean.view(shape)) / std.view(shape)
+    mean = _cached_const(KITTI_MEAN, x.device, x.dtype, shape)
+    std = _cached_const(KITTI_STD_SIFINDER, x.device, x.dtype, shape)
+    return (x - mean) / std
 
 
 def gaussian_mask_value(num_patches_w: int, ph: int, pw: int, H: int, W: int,
