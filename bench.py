@@ -56,6 +56,8 @@ def main():
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph capture of the train step")
+    ap.add_argument("--channels-last", action="store_true",
+                    help="NHWC memory format for the conv stack")
     args = ap.parse_args()
 
     local_rank = init_distributed()
@@ -70,6 +72,11 @@ def main():
     ae_config, pc_config = build_configs(args)
     torch.manual_seed(1234 + rank())
     model = DSIN(ae_config, pc_config).to(device)
+    if args.channels_last:
+        model.encoder = model.encoder.to(memory_format=torch.channels_last)
+        model.decoder = model.decoder.to(memory_format=torch.channels_last)
+        if model.sinet is not None:
+            model.sinet = model.sinet.to(memory_format=torch.channels_last)
     trainer = Trainer(model, ae_config, pc_config, num_training_imgs=1576,
                      device=device, autocast_bf16=(args.dtype == "bf16"),
                      use_cuda_graph=(not args.no_graph and device.type == "cuda"),
