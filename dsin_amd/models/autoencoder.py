@@ -43,23 +43,40 @@ def _bn(ch: int) -> nn.BatchNorm2d:
 
 
 class ConvBNAct(nn.Module):
+    """conv (or transpose conv) + BN + optional ReLU. The conv runs on the
+    custom gather-GEMM MFMA kernel on GPU (ops/conv.py) and torch eager on
+    CPU; nn.Conv2d modules are kept purely as parameter holders so the
+    state-dict layout stays conventional."""
+
     def __init__(self, cin: int, cout: int, k: int, stride: int = 1,
                  relu: bool = True, transpose: bool = False):
         super().__init__()
+        self.transpose = transpose
+        self.stride = stride
+        self.padding = (k - 1) // 2
+        self.output_padding = stride - 1 if transpose else 0
         if transpose:
             # out = 2*in for stride 2: pad=(k-1)//2, output_padding=1
             self.conv = nn.ConvTranspose2d(cin, cout, k, stride=stride,
-                                           padding=(k - 1) // 2, output_padding=stride - 1,
+                                           padding=self.padding,
+                                           output_padding=self.output_padding,
                                            bias=False)
         else:
-            self.conv = nn.Conv2d(cin, cout, k, stride=stride, padding=(k - 1) // 2,
-                                  bias=False)
+            self.conv = nn.Conv2d(cin, cout, k, stride=stride,
+                                  padding=self.padding, bias=False)
         self.bn = _bn(cout)
         self.relu = relu
 
     def forward(self, x):
-        x = self.bn(self.conv(x))
-        return torch.relu(x) if self.relu else x
+        from ..ops import conv as dconv
+        if self.transpose:
+            y = dconv.conv_transpose2d(x, self.conv.weight, None, self.stride,
+                                       self.padding, self.output_padding)
+        else:
+            y = dconv.conv2d(x, self.conv.weight, None, self.stride,
+                             self.padding, 1)
+        y = self.bn(y)
+        return torch.relu(y) if self.relu else y
 
 
 class ResidualBlock(nn.Module):

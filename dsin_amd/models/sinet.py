@@ -44,6 +44,10 @@ class SiNet(nn.Module):
         nn.init.zeros_(self.last.bias)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops import conv as dconv
         for conv in self.convs:
-            x = F.leaky_relu(conv(x), negative_slope=0.2)
-        return self.last(x)
+            d = conv.dilation[0]
+            # fused bias + leaky-relu epilogue on the gather-GEMM kernel
+            x = dconv.conv2d(x, conv.weight, conv.bias, stride=1, padding=d,
+                             dilation=d, act=2)
+        return dconv.conv2d(x, self.last.weight, self.last.bias, 1, 0, 1, act=0)

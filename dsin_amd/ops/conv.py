@@ -1,0 +1,169 @@
+"""Custom implicit-GEMM convolutions (gather-GEMM on MFMA, gfx950).
+
+Covers every 2D conv in DSIN: direct conv with any
+stride/dilation, transposed conv, and the full backward — via ONE device
+kernel pair (conv_fwd gather-GEMM + conv_wrw) plus torch-differentiable
+geometry transforms:
+
+  * forward pads (and zero-stuffs, for conv-transpose) the input into a
+    buffer, flattens the weight to (Co, K=(ci,r,s)); the kernel computes
+    out[pixel][cout] = sum_k xbuf[mbase[pixel]+koff[k]] * w[cout][k];
+  * backward-data is ANOTHER gather conv: stride-1 dilated conv of the
+    zero-stuffed padded dy with spatially-rotated ci<->co-swapped weights;
+  * backward-weight is the conv_wrw kernel (fp32 atomics over pixel chunks);
+  * pad / stuff / flip / permute happen in torch, so their gradients
+    (cropping, un-stuffing) come from autograd for free.
+
+bf16 compute with fp32 accumulate; fp32 master weights. Optional fused
+epilogue activation (relu / leaky-0.2) for the no-batchnorm convs (siNet).
+CPU path falls back to torch.nn.functional (the numerics oracle).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import _require_ext, hip_available
+
+_PLANS = {}
+
+
+def _plan(device, Ci: int, Hp: int, Wp: int, kh: int, kw: int, stride: int,
+          dil: int, HO: int, WO: int):
+    key = (device.index, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
+    p = _PLANS.get(key)
+    if p is None:
+        tables = _require_ext("conv_tables")
+        K = Ci * kh * kw
+        mbase, koff = tables(HO * WO, K, WO, stride, dil, Wp, Hp * Wp, kh, kw,
+                             device)
+        p = (mbase, koff)
+        _PLANS[key] = p
+    return p
+
+
+def _wmat(w1: torch.Tensor) -> torch.Tensor:
+    """(Co, K) any-dtype -> bf16 padded to (Co, KP+8)."""
+    K = w1.shape[1]
+    KP = (K + 31) & ~31
+    return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
+
+
+def _zero_stuff(x: torch.Tensor, stride: int) -> torch.Tensor:
+    if stride == 1:
+        return x
+    b, c, h, w = x.shape
+    z = x.new_zeros(b, c, (h - 1) * stride + 1, (w - 1) * stride + 1)
+    z[..., ::stride, ::stride] = x
+    return z
+
+
+class _GatherConvFn(torch.autograd.Function):
+    """y[b, co, oh, ow] = sum_{ci,r,s} xbuf[b, ci, oh*st + r*d, ow*st + s*d]
+    * w1[co, (ci, r, s)]  (+bias, +activation). VALID over xbuf."""
+
+    @staticmethod
+    def forward(ctx, xbuf, w1, bias, stride, dil, kh, kw, HO, WO, act):
+        ext_fwd = _require_ext("conv_fwd")
+        B, Ci, Hp, Wp = xbuf.shape
+        Co, K = w1.shape
+        mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
+        bias32 = bias.float().contiguous() if bias is not None else None
+        y = ext_fwd(xbuf, _wmat(w1), bias32, mbase, koff, Co, K, HO, WO, act)
+        ctx.save_for_backward(xbuf, w1, y if act else None)
+        ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        xbuf, w1, y_act = ctx.saved_tensors
+        stride, dil, kh, kw, HO, WO, has_bias, act = ctx.meta
+        ext_fwd = _require_ext("conv_fwd")
+        ext_wrw = _require_ext("conv_wrw")
+        B, Ci, Hp, Wp = xbuf.shape
+        Co, K = w1.shape
+
+        dy = dy.contiguous().to(torch.bfloat16)
+        if act == 1:
+            dy = torch.where(y_act > 0, dy, torch.zeros((), dtype=dy.dtype,
+                                                        device=dy.device))
+        elif act == 2:
+            dy = torch.where(y_act > 0, dy, dy * 0.2)
+
+        dxbuf = None
+        if ctx.needs_input_grad[0]:
+            with torch.no_grad():
+                z = _zero_stuff(dy, stride)
+                pe_h, pe_w = (kh - 1) * dil, (kw - 1) * dil
+                dybuf = F.pad(z, (pe_w, pe_w, pe_h, pe_h)).contiguous()
+                wrot = (w1.view(Co, Ci, kh, kw).flip(2, 3).permute(1, 0, 2, 3)
+                        .reshape(Ci, Co * kh * kw))
+                mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
+                                 kh, kw, 1, dil, Hp, Wp)
+                dxbuf = ext_fwd(dybuf, _wmat(wrot), None, mb2, ko2, Ci,
+                                Co * kh * kw, Hp, Wp, 0)
+
+        dw1 = None
+        if ctx.needs_input_grad[1]:
+            mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil,
+                                HO, WO)
+            dw1 = ext_wrw(xbuf, dy, mbase, koff, Co, K).to(w1.dtype)
+
+        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        return dxbuf, dw1, dbias, None, None, None, None, None, None, None
+
+
+def _out_size(h: int, k: int, s: int, p: int, d: int) -> int:
+    return (h + 2 * p - (k - 1) * d - 1) // s + 1
+
+
+def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
+           stride: int = 1, padding: int = 0, dilation: int = 1,
+           act: int = 0) -> torch.Tensor:
+    """Drop-in conv (NCHW). GPU: gather-GEMM MFMA kernel; CPU: torch oracle."""
+    if not x.is_cuda:
+        y = F.conv2d(x, w.to(x.dtype), bias.to(x.dtype) if bias is not None else None,
+                     stride=stride, padding=padding, dilation=dilation)
+        return _act(y, act)
+    B, Ci, H, W = x.shape
+    Co, _, kh, kw = w.shape
+    HO = _out_size(H, kh, stride, padding, dilation)
+    WO = _out_size(W, kw, stride, padding, dilation)
+    xbuf = F.pad(x.to(torch.bfloat16),
+                 (padding, padding, padding, padding)).contiguous()
+    return _GatherConvFn.apply(xbuf, w.reshape(Co, Ci * kh * kw), bias,
+                               stride, dilation, kh, kw, HO, WO, act)
+
+
+def conv_transpose2d(x: torch.Tensor, w: torch.Tensor,
+                     bias: Optional[torch.Tensor] = None, stride: int = 2,
+                     padding: int = 0, output_padding: int = 0,
+                     act: int = 0) -> torch.Tensor:
+    """Transposed conv (w: (Ci, Co, kh, kw), the torch layout)."""
+    if not x.is_cuda:
+        y = F.conv_transpose2d(x, w.to(x.dtype),
+                               bias.to(x.dtype) if bias is not None else None,
+                               stride=stride, padding=padding,
+                               output_padding=output_padding)
+        return _act(y, act)
+    B, Ci, H, W = x.shape
+    _, Co, kh, kw = w.shape
+    HO = (H - 1) * stride - 2 * padding + kh + output_padding
+    WO = (W - 1) * stride - 2 * padding + kw + output_padding
+    z = _zero_stuff(x.to(torch.bfloat16), stride)
+    pl_h, pl_w = kh - 1 - padding, kw - 1 - padding
+    xbuf = F.pad(z, (pl_w, pl_w + output_padding,
+                     pl_h, pl_h + output_padding)).contiguous()
+    w1 = w.flip(2, 3).permute(1, 0, 2, 3).reshape(Co, Ci * kh * kw)
+    return _GatherConvFn.apply(xbuf, w1, bias, 1, 1, kh, kw, HO, WO, act)
+
+
+def _act(y: torch.Tensor, act: int) -> torch.Tensor:
+    if act == 1:
+        return torch.relu(y)
+    if act == 2:
+        return F.leaky_relu(y, 0.2)
+    return y

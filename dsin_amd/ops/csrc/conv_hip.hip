@@ -1,0 +1,76 @@
+#include "hip/hip_runtime.h"
+// Host wrappers for the implicit-GEMM conv family (device code in
+// conv_kernels.h). See dsin_amd/ops/conv.py for the autograd layer and the
+// geometry-to-table mapping.
+
+#include "common_hip.h"
+#include "conv_kernels.h"
+
+namespace dsin {
+
+std::tuple<torch::Tensor, torch::Tensor> conv_tables(
+    int64_t M, int64_t K, int64_t WO, int64_t stride, int64_t dil, int64_t Wp,
+    int64_t HpWp, int64_t kh, int64_t kw, torch::Device device) {
+  auto opts = torch::TensorOptions().device(device).dtype(torch::kInt32);
+  auto mbase = torch::empty({M}, opts);
+  auto koff = torch::empty({K}, opts);
+  int64_t n = std::max(M, K);
+  hipLaunchKernelGGL(conv_tables_kernel, grid1d(n, 256), dim3(256), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     mbase.data_ptr<int>(), koff.data_ptr<int>(), (int)M,
+                     (int)K, (int)WO, (int)stride, (int)dil, (int)Wp,
+                     (int)HpWp, (int)(kh * kw), (int)kw);
+  return {mbase, koff};
+}
+
+torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
+                       c10::optional<torch::Tensor> bias,
+                       torch::Tensor mbase, torch::Tensor koff, int64_t N,
+                       int64_t K, int64_t HO, int64_t WO, int64_t act) {
+  CHECK_CUDA_CONTIG(xbuf);
+  CHECK_CUDA_CONTIG(wmat);
+  TORCH_CHECK(xbuf.scalar_type() == torch::kBFloat16, "xbuf must be bf16");
+  TORCH_CHECK(wmat.scalar_type() == torch::kBFloat16, "wmat must be bf16");
+  const int64_t B = xbuf.size(0);
+  const int64_t M = HO * WO;
+  const int64_t KP = (K + 31) & ~31;
+  TORCH_CHECK(wmat.size(1) == KP + CONV_AP, "wmat row stride mismatch");
+  auto out = torch::empty({B, N, HO, WO}, xbuf.options());
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_CUDA_CONTIG(bias.value());
+    bptr = bias->data_ptr<float>();
+  }
+  dim3 grid((M + CONV_TM - 1) / CONV_TM, (N + CONV_TN - 1) / CONV_TN, B);
+  size_t lds = (size_t)CONV_TM * (CONV_KC + CONV_AP) * 2;
+  hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(256), lds,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     (const cvbf16*)xbuf.data_ptr(),
+                     (const cvbf16*)wmat.data_ptr(), bptr,
+                     (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
+                     koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
+                     xbuf.stride(0), (long long)N * M, (int)act);
+  return out;
+}
+
+torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
+                       torch::Tensor mbase, torch::Tensor koff, int64_t N,
+                       int64_t K) {
+  CHECK_CUDA_CONTIG(xbuf);
+  CHECK_CUDA_CONTIG(dy);
+  const int64_t B = xbuf.size(0);
+  const int64_t M = dy.size(2) * dy.size(3);
+  auto dw = torch::zeros({N, K}, xbuf.options().dtype(torch::kFloat32));
+  int pix_chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 2048, 1), 64);
+  dim3 grid((K + 63) / 64, (N + 63) / 64, B * pix_chunks);
+  size_t lds = (size_t)2 * 64 * (32 + CONV_AP) * 2;
+  hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), lds,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                     (const cvbf16*)xbuf.data_ptr(),
+                     (const cvbf16*)dy.data_ptr(), dw.data_ptr<float>(),
+                     mbase.data_ptr<int>(), koff.data_ptr<int>(), (int)M,
+                     (int)N, (int)K, xbuf.stride(0), dy.stride(0), pix_chunks);
+  return dw;
+}
+
+}  // namespace dsin

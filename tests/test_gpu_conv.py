@@ -1,0 +1,126 @@
+"""Gather-GEMM conv kernel numerics vs torch fp32 oracles (MI355X)."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+from dsin_amd.ops import conv as dconv
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device("cuda:0")
+
+
+def _cmp_conv(dev, Ci, Co, H, W, k, stride=1, padding=0, dilation=1,
+              bias=False, act=0, atol=0.15, seed=0):
+    torch.manual_seed(seed)
+    x = torch.randn(2, Ci, H, W, device=dev)
+    w = torch.randn(Co, Ci, k, k, device=dev) / (k * Ci ** 0.5)
+    b = torch.randn(Co, device=dev) if bias else None
+    x1 = x.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    b1 = b.clone().requires_grad_(True) if bias else None
+    y = dconv.conv2d(x1, w1, b1, stride, padding, dilation, act)
+
+    x2 = x.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    b2 = b.clone().requires_grad_(True) if bias else None
+    yr = F.conv2d(x2, w2, b2, stride=stride, padding=padding, dilation=dilation)
+    if act == 1:
+        yr = torch.relu(yr)
+    elif act == 2:
+        yr = F.leaky_relu(yr, 0.2)
+
+    assert y.shape == yr.shape, (y.shape, yr.shape)
+    torch.testing.assert_close(y.float(), yr, rtol=0.05, atol=atol)
+
+    g = torch.randn_like(yr)
+    y.backward(g.to(y.dtype))
+    yr.backward(g)
+    torch.testing.assert_close(x1.grad.float(), x2.grad, rtol=0.08,
+                               atol=atol * 3)
+    torch.testing.assert_close(w1.grad.float(), w2.grad, rtol=0.08,
+                               atol=atol * 3)
+    if bias:
+        torch.testing.assert_close(b1.grad.float(), b2.grad, rtol=0.05,
+                                   atol=atol)
+
+
+def test_conv3x3_s1(dev):
+    _cmp_conv(dev, 32, 48, 24, 40, 3, stride=1, padding=1)
+
+
+def test_conv3x3_s1_128ch(dev):
+    _cmp_conv(dev, 128, 128, 20, 48, 3, stride=1, padding=1, atol=0.4)
+
+
+def test_conv5x5_s2(dev):
+    _cmp_conv(dev, 3, 64, 64, 96, 5, stride=2, padding=2)
+
+
+def test_conv5x5_s2_33(dev):
+    _cmp_conv(dev, 64, 33, 40, 48, 5, stride=2, padding=2, atol=0.4)
+
+
+def test_conv_dilated(dev):
+    _cmp_conv(dev, 6, 32, 40, 56, 3, stride=1, padding=4, dilation=4,
+              bias=True, act=2)
+
+
+def test_conv_dilated_large_rate(dev):
+    _cmp_conv(dev, 32, 32, 48, 72, 3, stride=1, padding=16, dilation=16,
+              bias=True, act=2)
+
+
+def test_conv1x1(dev):
+    _cmp_conv(dev, 32, 3, 24, 40, 1, stride=1, padding=0, bias=True)
+
+
+def _cmp_convT(dev, Ci, Co, H, W, k, stride=2, padding=None, seed=1,
+               atol=0.15):
+    if padding is None:
+        padding = (k - 1) // 2
+    torch.manual_seed(seed)
+    x = torch.randn(1, Ci, H, W, device=dev)
+    w = torch.randn(Ci, Co, k, k, device=dev) / (k * Ci ** 0.5)
+    x1 = x.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    y = dconv.conv_transpose2d(x1, w1, None, stride, padding, stride - 1)
+
+    x2 = x.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    yr = F.conv_transpose2d(x2, w2, None, stride=stride, padding=padding,
+                            output_padding=stride - 1)
+    assert y.shape == yr.shape, (y.shape, yr.shape)
+    torch.testing.assert_close(y.float(), yr, rtol=0.05, atol=atol)
+    g = torch.randn_like(yr)
+    y.backward(g.to(y.dtype))
+    yr.backward(g)
+    torch.testing.assert_close(x1.grad.float(), x2.grad, rtol=0.08, atol=atol * 3)
+    torch.testing.assert_close(w1.grad.float(), w2.grad, rtol=0.08, atol=atol * 3)
+
+
+def test_convT3x3_s2(dev):
+    _cmp_convT(dev, 32, 64, 8, 12, 3)
+
+
+def test_convT5x5_s2(dev):
+    _cmp_convT(dev, 64, 32, 16, 24, 5)
+
+
+def test_convT5x5_s2_to3(dev):
+    _cmp_convT(dev, 32, 3, 16, 24, 5)
+
+
+def test_conv_grad_isolation_first_layer(dev):
+    """x without requires_grad must skip bwd-data cleanly."""
+    torch.manual_seed(2)
+    x = torch.randn(1, 3, 32, 48, device=dev)
+    w = torch.randn(16, 3, 5, 5, device=dev, requires_grad=True)
+    y = dconv.conv2d(x, w, None, 2, 2, 1)
+    y.sum().backward()
+    assert w.grad is not None and torch.isfinite(w.grad).all()
