@@ -72,7 +72,8 @@ class _GatherConvFn(torch.autograd.Function):
         Co, K = w1.shape
         mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
         bias32 = bias.float().contiguous() if bias is not None else None
-        y = ext_fwd(xbuf, _wmat(w1), bias32, mbase, koff, Co, K, HO, WO, act)
+        y = ext_fwd(xbuf, _wmat(w1), bias32, mbase, koff, Co, K, HO, WO, act,
+                    stride == 1)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act)
         return y
@@ -104,13 +105,14 @@ class _GatherConvFn(torch.autograd.Function):
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
                                  kh, kw, 1, dil, Hp, Wp)
                 dxbuf = ext_fwd(dybuf, _wmat(wrot), None, mb2, ko2, Ci,
-                                Co * kh * kw, Hp, Wp, 0)
+                                Co * kh * kw, Hp, Wp, 0, True)
 
         dw1 = None
         if ctx.needs_input_grad[1]:
             mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil,
                                 HO, WO)
-            dw1 = ext_wrw(xbuf, dy, mbase, koff, Co, K).to(w1.dtype)
+            dw1 = ext_wrw(xbuf, dy, mbase, koff, Co, K, WO,
+                          stride == 1).to(w1.dtype)
 
         dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
         return dxbuf, dw1, dbias, None, None, None, None, None, None, None

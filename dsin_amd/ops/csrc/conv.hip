@@ -25,7 +25,8 @@ std::tuple<torch::Tensor, torch::Tensor> conv_tables(
 torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        c10::optional<torch::Tensor> bias,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
-                       int64_t K, int64_t HO, int64_t WO, int64_t act) {
+                       int64_t K, int64_t HO, int64_t WO, int64_t act,
+                       bool mcontig) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(wmat);
   TORCH_CHECK(xbuf.scalar_type() == torch::kBFloat16, "xbuf must be bf16");
@@ -41,20 +42,21 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
     bptr = bias->data_ptr<float>();
   }
   dim3 grid((M + CONV_TM - 1) / CONV_TM, (N + CONV_TN - 1) / CONV_TN, B);
-  size_t lds = (size_t)CONV_TM * (CONV_KC + CONV_AP) * 2;
+  size_t lds = (size_t)2 * CONV_TM * (CONV_KC + CONV_AP) * 2;  // double buffer
   hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(256), lds,
                      at::cuda::getCurrentCUDAStream(),
                      (const cvbf16*)xbuf.data_ptr(),
                      (const cvbf16*)wmat.data_ptr(), bptr,
                      (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
                      koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
-                     xbuf.stride(0), (long long)N * M, (int)act);
+                     xbuf.stride(0), (long long)N * M, (int)act, (int)WO,
+                     (int)(mcontig ? 1 : 0));
   return out;
 }
 
 torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
-                       int64_t K) {
+                       int64_t K, int64_t WO, bool mcontig) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(dy);
   const int64_t B = xbuf.size(0);
@@ -62,13 +64,14 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
   auto dw = torch::zeros({N, K}, xbuf.options().dtype(torch::kFloat32));
   int pix_chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 2048, 1), 64);
   dim3 grid((K + 63) / 64, (N + 63) / 64, B * pix_chunks);
-  size_t lds = (size_t)2 * 64 * (32 + CONV_AP) * 2;
+  size_t lds = (size_t)4 * 64 * (32 + CONV_AP) * 2;  // 2 tiles x dbuf
   hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), lds,
                      at::cuda::getCurrentCUDAStream(),
                      (const cvbf16*)xbuf.data_ptr(),
                      (const cvbf16*)dy.data_ptr(), dw.data_ptr<float>(),
                      mbase.data_ptr<int>(), koff.data_ptr<int>(), (int)M,
-                     (int)N, (int)K, xbuf.stride(0), dy.stride(0), pix_chunks);
+                     (int)N, (int)K, xbuf.stride(0), dy.stride(0), pix_chunks,
+                     (int)WO, (int)(mcontig ? 1 : 0));
   return dw;
 }
 

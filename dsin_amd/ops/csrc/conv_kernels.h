@@ -69,11 +69,13 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
                      int M, int N, int K, int KP,
                      long long x_img_stride,            // Ci*Hp*Wp
                      long long o_img_stride,            // Co*M
-                     int act) {                         // 0 none 1 relu 2 lrelu
+                     int act, int WO,
+                     int mcontig) {  // 1: mbase[m+1]==mbase[m]+1 within rows
   const int WSTRIDE = KP + CONV_AP;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  cvbf16* As = reinterpret_cast<cvbf16*>(smem);  // [CONV_TM][CONV_KC+CONV_AP]
+  cvbf16* As = reinterpret_cast<cvbf16*>(smem);  // 2 x [TM][KC+AP] dbuf
   const int ASTR = CONV_KC + CONV_AP;
+  const int ABUF = CONV_TM * ASTR;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -89,35 +91,66 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   cv_f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
                      {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 
-  // stage loop index precompute: element (m, k) of the current chunk
-  const int sm = tid & 63;          // pixel within tile (coalesced dim)
-  const int sk0 = tid >> 6;         // k within chunk, step 4
-  const int gm = m0 + sm;
-  const int mb = (gm < M) ? mbase[gm] : 0;
+  // staging assignment: thread t -> 8 consecutive pixels (sm8..sm8+7) at one
+  // k; vector global load when those pixels are address-contiguous
+  const int sm8 = (tid & 7) * 8;
+  const int sk = tid >> 3;                 // 0..31 (k within chunk)
+  const int gmv = m0 + sm8;
 
-  const int ncol = n0 + colL;       // this lane's output channel
+  const int ncol = n0 + colL;
   const cvbf16* wrow = wmat + (long long)(ncol < N ? ncol : 0) * WSTRIDE;
 
-  for (int kc = 0; kc < KP; kc += CONV_KC) {
-    // ---- stage A chunk (64 x 32): thread (sm, sk0+4t) ----
-    __syncthreads();
+  typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+  u16x8 stage;                             // staged values, 8 bf16
+  // contiguity needs: stride 1 (mcontig), in-range, and no output-row
+  // boundary inside the 8-pixel run (mbase jumps by Wp-WO+1 across rows)
+  const bool vecok = mcontig && (gmv + 7 < M) && ((gmv % WO) + 8 <= WO);
+
+  auto load_chunk = [&](int kc) {
+    const int k = kc + sk;
+    const int ko = (k < K) ? koff[k] : 0;
+    if (k < K && vecok) {
+      stage = *reinterpret_cast<const u16x8*>(&x[mbase[gmv] + ko]);
+    } else {
 #pragma unroll
-    for (int t = 0; t < CONV_KC / 4; ++t) {
-      const int k = kc + sk0 + 4 * t;
-      cvbf16 v = cvf2b(0.f);
-      if (k < K && gm < M) v = x[mb + koff[k]];
-      As[sm * ASTR + sk0 + 4 * t] = v;
+      for (int i = 0; i < 8; ++i) {
+        cvbf16 v = cvf2b(0.f);
+        const int gm = gmv + i;
+        if (k < K && gm < M) v = x[mbase[gm] + ko];
+        stage[i] = *reinterpret_cast<unsigned short*>(&v);
+      }
     }
-    __syncthreads();
-    // ---- MFMA: 4 row-subtiles x K=32 ----
+  };
+  auto write_chunk = [&](int buf) {
+    cvbf16* dst = As + buf * ABUF;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      unsigned short u = stage[i];
+      dst[(sm8 + i) * ASTR + sk] = *reinterpret_cast<cvbf16*>(&u);
+    }
+  };
+
+  load_chunk(0);
+  write_chunk(0);
+  __syncthreads();
+
+  const int nchunks = KP / CONV_KC;
+  for (int kt = 0; kt < nchunks; ++kt) {
+    // issue next chunk's global loads before computing the current one
+    if (kt + 1 < nchunks) load_chunk((kt + 1) * CONV_KC);
+    const cvbf16* cur = As + (kt & 1) * ABUF;
     const cv_bf16x8 bfrag = *reinterpret_cast<const cv_bf16x8*>(
-        &wrow[kc + kgrp * 8]);
+        &wrow[kt * CONV_KC + kgrp * 8]);
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
-          &As[(mi * 16 + colL) * ASTR + kgrp * 8]);
+          &cur[(mi * 16 + colL) * ASTR + kgrp * 8]);
       acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[mi],
                                                         0, 0, 0);
+    }
+    if (kt + 1 < nchunks) {
+      write_chunk((kt + 1) & 1);
+      __syncthreads();
     }
   }
 
@@ -152,23 +185,24 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
                      const int* __restrict__ koff,
                      int M, int N, int K,
                      long long x_img_stride, long long dy_img_stride,
-                     int pix_chunks) {
+                     int pix_chunks, int WO, int mcontig) {
   // blockIdx.x: tap tile; blockIdx.y: cout tile; blockIdx.z: pixel chunk*img
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  cvbf16* As = reinterpret_cast<cvbf16*>(smem);   // [64][32+AP] taps x pixels
-  cvbf16* Bs = As + 64 * (32 + CONV_AP);          // [64][32+AP] couts x pixels
   const int ASTR = 32 + CONV_AP;
+  const int TBUF = 64 * ASTR;
+  cvbf16* As = reinterpret_cast<cvbf16*>(smem);   // 2 x [64 taps][32 pixels]
+  cvbf16* Bs = As + 2 * TBUF;                     // 2 x [64 couts][32 pixels]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int k0 = blockIdx.x * 64;                 // tap tile base
-  const int n0 = blockIdx.y * 64 + wid * 16;      // cout tile base (per wave)
+  const int k0 = blockIdx.x * 64;
   const int img = blockIdx.z / pix_chunks;
   const int pc = blockIdx.z % pix_chunks;
-  const int PCHUNK = (M + pix_chunks - 1) / pix_chunks;
+  const int PCHUNK = ((M + pix_chunks - 1) / pix_chunks + 31) & ~31;
   const int p0 = pc * PCHUNK;
   const int p1 = min(p0 + PCHUNK, M);
+  if (p0 >= M) return;
 
   const cvbf16* x = xpad + (long long)img * x_img_stride;
   const cvbf16* g = dy + (long long)img * dy_img_stride;
@@ -179,40 +213,74 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
   cv_f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
                      {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 
-  const int sj = tid & 63;        // row of the staged tile (tap or cout)
-  const int sp0 = tid >> 6;       // pixel in chunk, step 4
+  // staging: thread t stages 8 consecutive pixels of row sj (a tap row of As
+  // and a cout row of Bs in alternating quads)
+  const int sj = tid >> 2;                        // 0..63 row
+  const int sp8 = (tid & 3) * 8;                  // pixel offset 0/8/16/24
   const int tap = k0 + sj;
   const int ko = (tap < K) ? koff[tap] : 0;
   const int cout = blockIdx.y * 64 + sj;
+  const long long gofs = (long long)(cout < N ? cout : 0) * M;
 
-  for (int pp = p0; pp < p1; pp += 32) {
-    __syncthreads();
+  typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+  u16x8 sa, sb;
+
+  auto load_chunk = [&](int pp) {
+    const int p = pp + sp8;
+    const bool inb = p + 7 < p1;
+    const bool avec = inb && mcontig && ((p % WO) + 8 <= WO);
+    if (avec && tap < K) {
+      sa = *reinterpret_cast<const u16x8*>(&x[mbase[p] + ko]);
+    } else {
 #pragma unroll
-    for (int t = 0; t < 8; ++t) {
-      const int p = pp + sp0 + 4 * t;
-      const bool pv = p < p1;
-      cvbf16 av = cvf2b(0.f), bvv = cvf2b(0.f);
-      if (pv && tap < K) av = x[mbase[p] + ko];
-      if (pv && cout < N) bvv = g[(long long)cout * M + p];
-      As[sj * ASTR + sp0 + 4 * t] = av;
-      Bs[sj * ASTR + sp0 + 4 * t] = bvv;
+      for (int i = 0; i < 8; ++i) {
+        cvbf16 v = cvf2b(0.f);
+        if (p + i < p1 && tap < K) v = x[mbase[p + i] + ko];
+        sa[i] = *reinterpret_cast<unsigned short*>(&v);
+      }
     }
-    __syncthreads();
+    if (inb && cout < N) {
+      sb = *reinterpret_cast<const u16x8*>(&g[gofs + p]);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        cvbf16 v = cvf2b(0.f);
+        if (p + i < p1 && cout < N) v = g[gofs + p + i];
+        sb[i] = *reinterpret_cast<unsigned short*>(&v);
+      }
+    }
+  };
+  auto write_chunk = [&](int buf) {
+    *reinterpret_cast<u16x8*>(&As[buf * TBUF + sj * ASTR + sp8]) = sa;
+    *reinterpret_cast<u16x8*>(&Bs[buf * TBUF + sj * ASTR + sp8]) = sb;
+  };
+
+  load_chunk(p0);
+  write_chunk(0);
+  __syncthreads();
+
+  const int nchunks = (p1 - p0 + 31) / 32;
+  for (int t = 0; t < nchunks; ++t) {
+    if (t + 1 < nchunks) load_chunk(p0 + (t + 1) * 32);
+    const cvbf16* ac = As + (t & 1) * TBUF;
+    const cvbf16* bc = Bs + (t & 1) * TBUF;
     const cv_bf16x8 bfrag = *reinterpret_cast<const cv_bf16x8*>(
-        &Bs[(n0 - blockIdx.y * 64 + colL) * ASTR + kgrp * 8]);
-    // NOTE: bfrag holds dy[cout=n0+colL][pixels kgrp*8..+7] — this is the
-    // MFMA B operand B[kdim=pixel][col=cout] fragment.
+        &bc[(wid * 16 + colL) * ASTR + kgrp * 8]);
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
-          &As[(mi * 16 + colL) * ASTR + kgrp * 8]);
+          &ac[(mi * 16 + colL) * ASTR + kgrp * 8]);
       acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[mi],
                                                         0, 0, 0);
+    }
+    if (t + 1 < nchunks) {
+      write_chunk((t + 1) & 1);
+      __syncthreads();
     }
   }
 
   // D[row=tap][col=cout]; accumulate into dw[cout][tap]
-  const int nc = n0 + colL;
+  const int nc = blockIdx.y * 64 + wid * 16 + colL;
   if (nc < N) {
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
