@@ -73,7 +73,7 @@ class _GatherConvFn(torch.autograd.Function):
         mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
         bias32 = bias.float().contiguous() if bias is not None else None
         y = ext_fwd(xbuf, _wmat(w1), bias32, mbase, koff, Co, K, HO, WO, act,
-                    stride == 1)
+                    stride)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act)
         return y
@@ -105,7 +105,7 @@ class _GatherConvFn(torch.autograd.Function):
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
                                  kh, kw, 1, dil, Hp, Wp)
                 dxbuf = ext_fwd(dybuf, _wmat(wrot), None, mb2, ko2, Ci,
-                                Co * kh * kw, Hp, Wp, 0, True)
+                                Co * kh * kw, Hp, Wp, 0, 1)
 
         dw1 = None
         if ctx.needs_input_grad[1]:

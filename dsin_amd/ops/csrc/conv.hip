@@ -26,7 +26,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        c10::optional<torch::Tensor> bias,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t HO, int64_t WO, int64_t act,
-                       bool mcontig) {
+                       int64_t stride) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(wmat);
   TORCH_CHECK(xbuf.scalar_type() == torch::kBFloat16, "xbuf must be bf16");
@@ -42,7 +42,8 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
     bptr = bias->data_ptr<float>();
   }
   dim3 grid((M + CONV_TM - 1) / CONV_TM, (N + CONV_TN - 1) / CONV_TN, B);
-  size_t lds = (size_t)2 * CONV_TM * (CONV_KC + CONV_AP) * 2;  // double buffer
+  const int64_t KPC = (K + 63) / 64 * 64;
+  size_t lds = (size_t)2 * CONV_TM * (64 + CONV_AP) * 2 + (size_t)KPC * 4;
   hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(256), lds,
                      at::cuda::getCurrentCUDAStream(),
                      (const cvbf16*)xbuf.data_ptr(),
@@ -50,7 +51,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                      (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
                      koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
                      xbuf.stride(0), (long long)N * M, (int)act, (int)WO,
-                     (int)(mcontig ? 1 : 0));
+                     (int)stride);
   return out;
 }
 

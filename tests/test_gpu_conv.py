@@ -43,11 +43,15 @@ def _cmp_conv(dev, Ci, Co, H, W, k, stride=1, padding=0, dilation=1,
     yr.backward(g)
     torch.testing.assert_close(x1.grad.float(), x2.grad, rtol=0.08,
                                atol=atol * 3)
+    # dw/db are sums over all pixels: scale the absolute tolerance with the
+    # gradient magnitude (bf16 relative error accumulates on large sums)
+    atol_w = max(atol * 3, 4e-3 * float(w2.grad.abs().max()))
     torch.testing.assert_close(w1.grad.float(), w2.grad, rtol=0.08,
-                               atol=atol * 3)
+                               atol=atol_w)
     if bias:
+        atol_b = max(atol, 4e-3 * float(b2.grad.abs().max()))
         torch.testing.assert_close(b1.grad.float(), b2.grad, rtol=0.05,
-                                   atol=atol)
+                                   atol=atol_b)
 
 
 def test_conv3x3_s1(dev):
