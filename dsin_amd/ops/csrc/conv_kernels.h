@@ -141,14 +141,25 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
     load_half(kc + sk, stage[0]);
     load_half(kc + 32 + sk, stage[1]);
   };
+  // A-tile byte-address XOR swizzle: staging writes hit banks at an
+  // m-stride of 8 rows (8*ASTR dwords = 0 mod 32 -> 8-way conflict);
+  // XOR-ing bits 4..6 with the row octave (m>>3) spreads them. Rows stay
+  // 16B-aligned and each 128B block stays inside one 8-row octave
+  // (8*ASTR*2 = 1152 B, a multiple of 128), so the map is a bijection and
+  // the b128 fragment reads just apply the same XOR.
+  auto aswz = [&](int m, int elem_off) -> int {
+    return ((m * ASTR + elem_off) * 2) ^ (((m >> 3) & 7) << 4);
+  };
+  char* As8 = reinterpret_cast<char*>(As);
   auto write_chunk = [&](int buf) {
-    cvbf16* dst = As + buf * ABUF;
+    char* dst = As8 + buf * ABUF * 2;
 #pragma unroll
     for (int h = 0; h < 2; ++h)
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         unsigned short u = stage[h][i];
-        dst[(sm8 + i) * ASTR + h * 32 + sk] = *reinterpret_cast<cvbf16*>(&u);
+        *reinterpret_cast<cvbf16*>(&dst[aswz(sm8 + i, h * 32 + sk)]) =
+            *reinterpret_cast<cvbf16*>(&u);
       }
   };
 
@@ -159,7 +170,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   const int nchunks = (K + KC - 1) / KC;
   for (int kt = 0; kt < nchunks; ++kt) {
     if (kt + 1 < nchunks) load_chunk((kt + 1) * KC);
-    const cvbf16* cur = As + (kt & 1) * ABUF;
+    const char* cur = As8 + (kt & 1) * ABUF * 2;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const int kb = kt * KC + kk * 32;
@@ -169,7 +180,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
-            &cur[(mi * 16 + colL) * ASTR + kk * 32 + kgrp * 8]);
+            &cur[aswz(mi * 16 + colL, kk * 32 + kgrp * 8)]);
         acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                           acc[mi], 0, 0, 0);
       }
