@@ -46,9 +46,10 @@ def _plan(device, Ci: int, Hp: int, Wp: int, kh: int, kw: int, stride: int,
 
 
 def _wmat(w1: torch.Tensor) -> torch.Tensor:
-    """(Co, K) any-dtype -> bf16 padded to (Co, KP+8)."""
+    """(Co, K) any-dtype -> bf16 zero-padded to (Co, KP64+8); the zeros
+    cancel the clamped out-of-range A gathers in the kernel."""
     K = w1.shape[1]
-    KP = (K + 31) & ~31
+    KP = (K + 63) & ~63
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
 
 

@@ -34,8 +34,9 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   TORCH_CHECK(wmat.scalar_type() == torch::kBFloat16, "wmat must be bf16");
   const int64_t B = xbuf.size(0);
   const int64_t M = HO * WO;
-  const int64_t KP = (K + 31) & ~31;
+  const int64_t KP = (K + 63) & ~63;  // 64-chunk padded; wmat zero-padded
   TORCH_CHECK(wmat.size(1) == KP + CONV_AP, "wmat row stride mismatch");
+  if (WO < 8) stride = 0;  // scalar staging path for very narrow outputs
   auto out = torch::empty({B, N, HO, WO}, xbuf.options());
   const float* bptr = nullptr;
   if (bias.has_value()) {
@@ -43,8 +44,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
     bptr = bias->data_ptr<float>();
   }
   dim3 grid((M + CONV_TM - 1) / CONV_TM, (N + CONV_TN - 1) / CONV_TN, B);
-  const int64_t KPC = (K + 63) / 64 * 64;
-  size_t lds = (size_t)2 * CONV_TM * (64 + CONV_AP) * 2 + (size_t)KPC * 4;
+  size_t lds = (size_t)2 * CONV_TM * (64 + CONV_AP) * 2 + (size_t)KP * 4;
   hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(256), lds,
                      at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                      (const cvbf16*)xbuf.data_ptr(),

@@ -61,24 +61,31 @@ __global__ void conv_tables_kernel(int* __restrict__ mbase,
 
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
-                     const cvbf16* __restrict__ wmat,   // (Co, KP+AP) padded
+                     const cvbf16* __restrict__ wmat,   // (Co, KP64+AP) 0-pad
                      const float* __restrict__ bias,    // (Co,) or nullptr
                      cvbf16* __restrict__ out,          // (Co, M) i.e. NCHW
                      const int* __restrict__ mbase,     // (M,)
                      const int* __restrict__ koff,      // (K,)
-                     int M, int N, int K, int KP,
+                     int M, int N, int K, int KP,       // KP: 64-multiple
                      long long x_img_stride,            // Ci*Hp*Wp
                      long long o_img_stride,            // Co*M
                      int act, int WO,
-                     int stride) {                      // for vector staging
-  // K chunk = 64 (two MFMA k-steps per barrier), double-buffered LDS.
+                     int stride) {
+  // Branchless pipeline: K chunks of 64 (two MFMA k-steps per barrier),
+  // double-buffered LDS A-tile with XOR-swizzled addressing. Staging is two
+  // overlapped 16B loads per 8-pixel run with a per-element crossing select
+  // (handles output-row boundaries without divergence); out-of-range filter
+  // taps read clamped addresses and are cancelled by the zero padding of
+  // wmat. W fragments for the NEXT chunk are prefetched to registers ahead
+  // of the A loads so the compiler's wait before the MFMAs is a counted
+  // vmcnt, not a pipeline-draining vmcnt(0).
   const int KC = 64;
   const int WSTRIDE = KP + CONV_AP;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int ASTR = KC + CONV_AP;
   const int ABUF = CONV_TM * ASTR;
   cvbf16* As = reinterpret_cast<cvbf16*>(smem);          // 2 x [TM][KC+AP]
-  int* Ko = reinterpret_cast<int*>(As + 2 * ABUF);       // [KPC] koff copy
+  int* Ko = reinterpret_cast<int*>(As + 2 * ABUF);       // [KP] koff copy
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -91,62 +98,72 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   const int colL = lane & 15;
   const int kgrp = lane >> 4;
 
-  const int KPC = (K + KC - 1) / KC * KC;
-  for (int k = tid; k < KPC; k += 256) Ko[k] = (k < K) ? koff[k] : -1;
+  for (int k = tid; k < KP; k += 256) Ko[k] = (k < K) ? koff[k] : 0;
 
   cv_f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
                      {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 
-  // staging: thread t -> 8 consecutive output pixels at k = (t>>3) and
-  // k+32 (two sub-chunks of the 64-wide chunk)
+  // staging: thread t -> 8 consecutive output pixels at k = (t>>3), k+32
   const int sm8 = (tid & 7) * 8;
   const int sk = tid >> 3;                 // 0..31
-  const int gmv = m0 + sm8;
-  const int mb0 = (gmv < M) ? mbase[gmv] : 0;
-  // contiguity of the 8-pixel run: no output-row boundary inside it
-  const bool rowok = (gmv + 7 < M) && ((gmv % WO) + 8 <= WO);
-  const bool vec1 = rowok && stride == 1;
-  const bool vec2 = rowok && stride == 2;
+  const int gm0 = min(m0 + sm8, M - 1);
+  const int gm7 = min(m0 + sm8 + 7, M - 1);
+  const int mb0 = mbase[gm0];
+  const int mb1 = mbase[gm7] - 7 * stride; // base so elem i sits at +i*stride
+  // crossing point: first i whose pixel falls on the next output row
+  const int cross = WO - (gm0 % WO);       // >= 8 means no crossing
+  const int xmax = (int)x_img_stride - 16; // clamp for safe 16B reads
 
   const int ncol = n0 + colL;
   const cvbf16* wrow = wmat + (long long)(ncol < N ? ncol : 0) * WSTRIDE;
 
   typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
   u16x8 stage[2];
+  cv_bf16x8 wfrag[2];
 
-  __syncthreads();  // Ko ready (also covers the first-stage write below)
+  __syncthreads();  // Ko ready
 
   auto load_half = [&](int k, u16x8& st) {
-    const int ko = (k < K) ? Ko[k] : -1;
-    if (ko >= 0 && vec1) {
-      st = *reinterpret_cast<const u16x8*>(&x[mb0 + ko]);
-    } else if (ko >= 0 && vec2 && mb0 + ko + 15 < (int)x_img_stride) {
-      // reads 16 elements to pick the 8 even (stride-2) ones; the guard
-      // keeps the 1-element overshoot inside this image's buffer
-      const u16x8 a = *reinterpret_cast<const u16x8*>(&x[mb0 + ko]);
-      const u16x8 b = *reinterpret_cast<const u16x8*>(&x[mb0 + ko + 8]);
+    const int ko = Ko[k];
+    if (stride == 1) {
+      const u16x8 a = *reinterpret_cast<const u16x8*>(
+          &x[min(mb0 + ko, xmax)]);
+      const u16x8 b = *reinterpret_cast<const u16x8*>(
+          &x[min(max(mb1 + ko, 0), xmax)]);
 #pragma unroll
-      for (int i = 0; i < 4; ++i) { st[i] = a[2 * i]; st[4 + i] = b[2 * i]; }
+      for (int i = 0; i < 8; ++i) st[i] = (i < cross) ? a[i] : b[i];
+    } else if (stride == 2) {
+      const int b0 = min(mb0 + ko, xmax);
+      const int b1 = min(max(mb1 + ko, 0), xmax);
+      const u16x8 a0 = *reinterpret_cast<const u16x8*>(&x[b0]);
+      const u16x8 a1 = *reinterpret_cast<const u16x8*>(&x[min(b0 + 8, xmax)]);
+      const u16x8 c0 = *reinterpret_cast<const u16x8*>(&x[b1]);
+      const u16x8 c1 = *reinterpret_cast<const u16x8*>(&x[min(b1 + 8, xmax)]);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        st[i] = (i < cross) ? a0[2 * i] : c0[2 * i];
+        st[4 + i] = (4 + i < cross) ? a1[2 * i] : c1[2 * i];
+      }
     } else {
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        cvbf16 v = cvf2b(0.f);
-        const int gm = gmv + i;
-        if (ko >= 0 && gm < M) v = x[mbase[gm] + ko];
+        const int gm = min(m0 + sm8 + i, M - 1);
+        cvbf16 v = x[min(mbase[gm] + ko, (int)x_img_stride - 1)];
         st[i] = *reinterpret_cast<unsigned short*>(&v);
       }
     }
   };
   auto load_chunk = [&](int kc) {
+    // W first: its consumer (the MFMA phase) waits on it with a counted
+    // vmcnt that leaves the later A loads in flight
+    wfrag[0] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + kgrp * 8]);
+    wfrag[1] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + 32 + kgrp * 8]);
     load_half(kc + sk, stage[0]);
     load_half(kc + 32 + sk, stage[1]);
   };
-  // A-tile byte-address XOR swizzle: staging writes hit banks at an
-  // m-stride of 8 rows (8*ASTR dwords = 0 mod 32 -> 8-way conflict);
-  // XOR-ing bits 4..6 with the row octave (m>>3) spreads them. Rows stay
-  // 16B-aligned and each 128B block stays inside one 8-row octave
-  // (8*ASTR*2 = 1152 B, a multiple of 128), so the map is a bijection and
-  // the b128 fragment reads just apply the same XOR.
+  // A-tile byte-address XOR swizzle (see write/read pair): staging writes at
+  // an 8-row stride collide on banks; rows stay 16B aligned and 128B blocks
+  // stay inside one 8-row octave (8*ASTR*2 = 1152 B) so the map is bijective
   auto aswz = [&](int m, int elem_off) -> int {
     return ((m * ASTR + elem_off) * 2) ^ (((m >> 3) & 7) << 4);
   };
@@ -165,18 +182,16 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 
   load_chunk(0);
   write_chunk(0);
+  cv_bf16x8 wcur[2] = {wfrag[0], wfrag[1]};
   __syncthreads();
 
-  const int nchunks = (K + KC - 1) / KC;
+  const int nchunks = KP / KC;
   for (int kt = 0; kt < nchunks; ++kt) {
     if (kt + 1 < nchunks) load_chunk((kt + 1) * KC);
     const char* cur = As8 + (kt & 1) * ABUF * 2;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      const int kb = kt * KC + kk * 32;
-      if (kb >= KP) break;
-      const cv_bf16x8 bfrag = *reinterpret_cast<const cv_bf16x8*>(
-          &wrow[kb + kgrp * 8]);
+      const cv_bf16x8 bfrag = wcur[kk];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
@@ -186,6 +201,8 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
       }
     }
     if (kt + 1 < nchunks) {
+      wcur[0] = wfrag[0];
+      wcur[1] = wfrag[1];
       write_chunk((kt + 1) & 1);
       __syncthreads();
     }
