@@ -53,6 +53,20 @@ def _wmat(w1: torch.Tensor) -> torch.Tensor:
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
 
 
+def _padded_buf(x: torch.Tensor, pl: int, pr: int, pt: int, pb: int
+                ) -> torch.Tensor:
+    """Zero-pad NCHW into a buffer with 16 elements of tail slack so the
+    kernel's 16B vector staging may overshoot reads without faulting
+    (garbage lanes are cancelled by weight zero-padding / output masking)."""
+    B, C, H, W = x.shape
+    Hp, Wp = H + pt + pb, W + pl + pr
+    n = C * Hp * Wp
+    store = x.new_zeros(B * n + 16)
+    buf = store[:B * n].view(B, C, Hp, Wp)
+    buf[:, :, pt:pt + H, pl:pl + W] = x
+    return buf
+
+
 def _zero_stuff(x: torch.Tensor, stride: int) -> torch.Tensor:
     if stride == 1:
         return x
@@ -100,7 +114,7 @@ class _GatherConvFn(torch.autograd.Function):
             with torch.no_grad():
                 z = _zero_stuff(dy, stride)
                 pe_h, pe_w = (kh - 1) * dil, (kw - 1) * dil
-                dybuf = F.pad(z, (pe_w, pe_w, pe_h, pe_h)).contiguous()
+                dybuf = _padded_buf(z, pe_w, pe_w, pe_h, pe_h)
                 wrot = (w1.view(Co, Ci, kh, kw).flip(2, 3).permute(1, 0, 2, 3)
                         .reshape(Ci, Co * kh * kw))
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
@@ -135,8 +149,8 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
     Co, _, kh, kw = w.shape
     HO = _out_size(H, kh, stride, padding, dilation)
     WO = _out_size(W, kw, stride, padding, dilation)
-    xbuf = F.pad(x.to(torch.bfloat16),
-                 (padding, padding, padding, padding)).contiguous()
+    xbuf = _padded_buf(x.to(torch.bfloat16), padding, padding, padding,
+                       padding)
     return _GatherConvFn.apply(xbuf, w.reshape(Co, Ci * kh * kw), bias,
                                stride, dilation, kh, kw, HO, WO, act)
 
@@ -158,8 +172,8 @@ def conv_transpose2d(x: torch.Tensor, w: torch.Tensor,
     WO = (W - 1) * stride - 2 * padding + kw + output_padding
     z = _zero_stuff(x.to(torch.bfloat16), stride)
     pl_h, pl_w = kh - 1 - padding, kw - 1 - padding
-    xbuf = F.pad(z, (pl_w, pl_w + output_padding,
-                     pl_h, pl_h + output_padding)).contiguous()
+    xbuf = _padded_buf(z, pl_w, pl_w + output_padding,
+                       pl_h, pl_h + output_padding)
     w1 = w.flip(2, 3).permute(1, 0, 2, 3).reshape(Co, Ci * kh * kw)
     return _GatherConvFn.apply(xbuf, w1, bias, 1, 1, kh, kw, HO, WO, act)
 

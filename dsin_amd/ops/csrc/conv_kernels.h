@@ -112,7 +112,8 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   const int mb1 = mbase[gm7] - 7 * stride; // base so elem i sits at +i*stride
   // crossing point: first i whose pixel falls on the next output row
   const int cross = WO - (gm0 % WO);       // >= 8 means no crossing
-  const int xmax = (int)x_img_stride - 16; // clamp for safe 16B reads
+  // xbuf carries >= 16 elements of tail slack (ops/conv.py _padded_buf), so
+  // in-range vector reads are never clamped; only negative bases are.
 
   const int ncol = n0 + colL;
   const cvbf16* wrow = wmat + (long long)(ncol < N ? ncol : 0) * WSTRIDE;
@@ -126,19 +127,18 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   auto load_half = [&](int k, u16x8& st) {
     const int ko = Ko[k];
     if (stride == 1) {
-      const u16x8 a = *reinterpret_cast<const u16x8*>(
-          &x[min(mb0 + ko, xmax)]);
+      const u16x8 a = *reinterpret_cast<const u16x8*>(&x[mb0 + ko]);
       const u16x8 b = *reinterpret_cast<const u16x8*>(
-          &x[min(max(mb1 + ko, 0), xmax)]);
+          &x[max(mb1 + ko, 0)]);
 #pragma unroll
       for (int i = 0; i < 8; ++i) st[i] = (i < cross) ? a[i] : b[i];
     } else if (stride == 2) {
-      const int b0 = min(mb0 + ko, xmax);
-      const int b1 = min(max(mb1 + ko, 0), xmax);
+      const int b0 = mb0 + ko;
+      const int b1 = max(mb1 + ko, 0);
       const u16x8 a0 = *reinterpret_cast<const u16x8*>(&x[b0]);
-      const u16x8 a1 = *reinterpret_cast<const u16x8*>(&x[min(b0 + 8, xmax)]);
+      const u16x8 a1 = *reinterpret_cast<const u16x8*>(&x[b0 + 8]);
       const u16x8 c0 = *reinterpret_cast<const u16x8*>(&x[b1]);
-      const u16x8 c1 = *reinterpret_cast<const u16x8*>(&x[min(b1 + 8, xmax)]);
+      const u16x8 c1 = *reinterpret_cast<const u16x8*>(&x[b1 + 8]);
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         st[i] = (i < cross) ? a0[2 * i] : c0[2 * i];
