@@ -85,7 +85,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   const int ASTR = KC + CONV_AP;
   const int ABUF = CONV_TM * ASTR;
   cvbf16* As = reinterpret_cast<cvbf16*>(smem);          // 2 x [TM][KC+AP]
-  int* Ko = reinterpret_cast<int*>(As + 2 * ABUF);       // [KP] koff copy
+  int* Ko = reinterpret_cast<int*>(As + 4 * ABUF);       // [KP] koff copy
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
