@@ -151,3 +151,19 @@ def test_conv_grad_isolation_first_layer(dev):
     y = dconv.conv2d(x, w, None, 2, 2, 1)
     y.sum().backward()
     assert w.grad is not None and torch.isfinite(w.grad).all()
+
+
+def test_conv_stress_repeated(dev):
+    """Repeated launches across the model's conv shapes: catches rare
+    race/fault behavior that single-shot tests miss."""
+    torch.manual_seed(0)
+    shapes = [(32, 48, 24, 40, 3, 1, 1), (128, 128, 40, 48, 3, 1, 1),
+              (3, 64, 64, 96, 5, 2, 2), (64, 33, 40, 48, 5, 2, 2)]
+    tensors = [(torch.randn(1, ci, h, w, device=dev),
+                torch.randn(co, ci, k, k, device=dev) * 0.05, st, p)
+               for ci, co, h, w, k, st, p in shapes]
+    for rep in range(50):
+        for x, w, st, p in tensors:
+            y = dconv.conv2d(x, w, None, st, p, 1)
+    torch.cuda.synchronize()
+    assert torch.isfinite(y.float()).all()
