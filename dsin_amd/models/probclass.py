@@ -62,8 +62,9 @@ class MaskedConv3d(nn.Module):
         self.bias = nn.Parameter(torch.zeros(cout))
         self.register_buffer("mask", mask.to(w.dtype), persistent=False)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.conv3d(x, self.weight * self.mask, self.bias)
+    def forward(self, x: torch.Tensor, act: int = 0) -> torch.Tensor:
+        from ..ops import conv as dconv
+        return dconv.conv3d_valid(x, self.weight * self.mask, self.bias, act)
 
 
 class ProbClass(nn.Module):
@@ -91,10 +92,11 @@ class ProbClass(nn.Module):
         return self.num_layers() * (self.K - 1) + 1  # reference :47-52
 
     def logits(self, q_pad: torch.Tensor) -> torch.Tensor:
-        """q_pad: (N, 1, D+pad, H+2pad, W+2pad) -> logits (N, L, D, H, W)."""
-        net = F.relu(self.conv0(q_pad))
+        """q_pad: (N, 1, D+pad, H+2pad, W+2pad) -> logits (N, L, D, H, W).
+        ReLUs fused into the masked-conv epilogue on GPU."""
+        net = self.conv0(q_pad, act=1)
         inner = net
-        net = F.relu(self.res_conv1(net))
+        net = self.res_conv1(net, act=1)
         net = self.res_conv2(net)
         net = net + inner[:, :, 2:, 2:-2, 2:-2]  # VALID shrink tracking (:196)
         return self.conv2(net)

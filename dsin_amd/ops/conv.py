@@ -184,3 +184,128 @@ def _act(y: torch.Tensor, act: int) -> torch.Tensor:
     if act == 2:
         return F.leaky_relu(y, 0.2)
     return y
+
+
+# ---------------------------------------------------------------------------
+# 3D conv on the same gather-GEMM kernels (the kernels are dimension-blind:
+# geometry lives in the offset tables). Used by the masked causal conv3d of
+# the entropy model (models/probclass.py) — input is pre-padded by
+# pad_for_probclass, the conv itself is VALID, stride 1.
+# ---------------------------------------------------------------------------
+
+_PLANS3D = {}
+
+
+def _plan3d(device, Ci, Dp, Hp, Wp, kd, kh, kw):
+    key = (device.index, Ci, Dp, Hp, Wp, kd, kh, kw)
+    p = _PLANS3D.get(key)
+    if p is None:
+        Do, Ho, Wo = Dp - kd + 1, Hp - kh + 1, Wp - kw + 1
+        m = torch.arange(Do * Ho * Wo, device=device, dtype=torch.int32)
+        od = torch.div(m, Ho * Wo, rounding_mode="floor")
+        rem = m % (Ho * Wo)
+        oh = torch.div(rem, Wo, rounding_mode="floor")
+        ow = rem % Wo
+        mbase = (od * (Hp * Wp) + oh * Wp + ow).contiguous()
+        k = torch.arange(Ci * kd * kh * kw, device=device, dtype=torch.int32)
+        khw = kd * kh * kw
+        ci = torch.div(k, khw, rounding_mode="floor")
+        rem = k % khw
+        r = torch.div(rem, kh * kw, rounding_mode="floor")
+        rem2 = rem % (kh * kw)
+        a = torch.div(rem2, kw, rounding_mode="floor")
+        b = rem2 % kw
+        koff = (ci * (Dp * Hp * Wp) + r * (Hp * Wp) + a * Wp + b).contiguous()
+        p = (mbase, koff, Do, Ho, Wo)
+        _PLANS3D[key] = p
+    return p
+
+
+class _GatherConv3dFn(torch.autograd.Function):
+    """VALID stride-1 3D conv: y[b,co,od,oh,ow] = sum xbuf[b,ci,od+r,oh+a,
+    ow+b'] * w1[co,(ci,r,a,b')] + bias, optional fused ReLU."""
+
+    @staticmethod
+    def forward(ctx, xbuf, w1, bias, kd, kh, kw, act):
+        ext_fwd = _require_ext("conv_fwd")
+        B, Ci, Dp, Hp, Wp = xbuf.shape
+        Co, K = w1.shape
+        mbase, koff, Do, Ho, Wo = _plan3d(xbuf.device, Ci, Dp, Hp, Wp, kd, kh, kw)
+        bias32 = bias.float().contiguous() if bias is not None else None
+        # kernel sees a 2D problem: M = Do*Ho*Wo pixels, "WO" = Wo rows
+        y = ext_fwd(xbuf.view(B, Ci, Dp * Hp, Wp), _wmat(w1), bias32, mbase,
+                    koff, Co, K, Do * Ho, Wo, act, 1)
+        ctx.save_for_backward(xbuf, w1, y if act else None)
+        ctx.meta = (kd, kh, kw, Do, Ho, Wo, bias is not None, act)
+        return y.view(B, Co, Do, Ho, Wo)
+
+    @staticmethod
+    def backward(ctx, dy):
+        xbuf, w1, y_act = ctx.saved_tensors
+        kd, kh, kw, Do, Ho, Wo, has_bias, act = ctx.meta
+        ext_fwd = _require_ext("conv_fwd")
+        ext_wrw = _require_ext("conv_wrw")
+        B, Ci, Dp, Hp, Wp = xbuf.shape
+        Co, K = w1.shape
+        dy = dy.contiguous().to(torch.bfloat16)
+        if act == 1:
+            dy = torch.where(y_act.view_as(dy) > 0, dy,
+                             torch.zeros((), dtype=dy.dtype, device=dy.device))
+
+        dxbuf = None
+        if ctx.needs_input_grad[0]:
+            with torch.no_grad():
+                dybuf = _padded_buf3d(dy, kd - 1, kh - 1, kw - 1)
+                wrot = (w1.view(Co, Ci, kd, kh, kw).flip(2, 3, 4)
+                        .permute(1, 0, 2, 3, 4).reshape(Ci, Co * kd * kh * kw))
+                mb2, ko2, D2, H2, W2 = _plan3d(dy.device, Co, dybuf.shape[2],
+                                               dybuf.shape[3], dybuf.shape[4],
+                                               kd, kh, kw)
+                assert (D2, H2, W2) == (Dp, Hp, Wp)
+                dxbuf = ext_fwd(dybuf.view(B, Co, -1, dybuf.shape[4]),
+                                _wmat(wrot), None, mb2, ko2, Ci,
+                                Co * kd * kh * kw, Dp * Hp, Wp, 0,
+                                1).view(B, Ci, Dp, Hp, Wp)
+
+        dw1 = None
+        if ctx.needs_input_grad[1]:
+            mbase, koff, _, _, _ = _plan3d(xbuf.device, Ci, Dp, Hp, Wp,
+                                           kd, kh, kw)
+            dw1 = ext_wrw(xbuf.view(B, Ci, Dp * Hp, Wp),
+                          dy.view(B, Co, Do * Ho, Wo), mbase, koff, Co, K,
+                          Wo, True).to(w1.dtype)
+
+        dbias = dy.float().sum(dim=(0, 2, 3, 4)) if has_bias else None
+        return dxbuf, dw1, dbias, None, None, None, None
+
+
+def _padded_buf3d(x: torch.Tensor, pd: int, ph: int, pw: int) -> torch.Tensor:
+    """Symmetric-pad NCDHW (all-sides pd/ph/pw) with 16-element tail slack."""
+    B, C, D, H, W = x.shape
+    Dp, Hp, Wp = D + 2 * pd, H + 2 * ph, W + 2 * pw
+    n = C * Dp * Hp * Wp
+    store = x.new_zeros(B * n + 16)
+    buf = store[:B * n].view(B, C, Dp, Hp, Wp)
+    buf[:, :, pd:pd + D, ph:ph + H, pw:pw + W] = x
+    return buf
+
+
+def conv3d_valid(x: torch.Tensor, w: torch.Tensor,
+                 bias: Optional[torch.Tensor] = None, act: int = 0
+                 ) -> torch.Tensor:
+    """VALID stride-1 3D conv; w: (Co, Ci, kD, kH, kW). GPU: gather-GEMM
+    kernels; CPU: torch oracle. Input must already carry any padding (the
+    probclass pads with centers[0] itself)."""
+    if not x.is_cuda:
+        y = F.conv3d(x, w.to(x.dtype),
+                     bias.to(x.dtype) if bias is not None else None)
+        return _act(y, act)
+    Co, Ci, kd, kh, kw = w.shape
+    # add the vector-staging slack (fresh buffer; conv3d inputs are small)
+    B = x.shape[0]
+    n = x[0].numel()
+    store = x.new_empty(B * n + 16, dtype=torch.bfloat16)
+    store[:B * n].copy_(x.reshape(-1))
+    xbuf = store[:B * n].view(x.shape)
+    return _GatherConv3dFn.apply(xbuf, w.reshape(Co, Ci * kd * kh * kw), bias,
+                                 kd, kh, kw, act)

@@ -167,3 +167,43 @@ def test_conv_stress_repeated(dev):
             y = dconv.conv2d(x, w, None, st, p, 1)
     torch.cuda.synchronize()
     assert torch.isfinite(y.float()).all()
+
+
+def test_conv3d_valid_vs_torch(dev):
+    torch.manual_seed(3)
+    x = torch.randn(1, 24, 34, 28, 36, device=dev)  # (B,Ci,D,H,W) padded-ish
+    w = torch.randn(24, 24, 2, 3, 3, device=dev) * 0.05
+    b = torch.randn(24, device=dev)
+    x1 = x.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    b1 = b.clone().requires_grad_(True)
+    y = dconv.conv3d_valid(x1, w1, b1, act=1)
+    x2 = x.clone().requires_grad_(True)
+    w2 = w.clone().requires_grad_(True)
+    b2 = b.clone().requires_grad_(True)
+    yr = torch.relu(F.conv3d(x2, w2, b2))
+    assert y.shape == yr.shape
+    torch.testing.assert_close(y.float(), yr, rtol=0.05, atol=0.2)
+    g = torch.randn_like(yr)
+    y.backward(g.to(y.dtype))
+    yr.backward(g)
+    torch.testing.assert_close(x1.grad.float(), x2.grad, rtol=0.08, atol=0.5)
+    atol_w = max(0.5, 4e-3 * float(w2.grad.abs().max()))
+    torch.testing.assert_close(w1.grad.float(), w2.grad, rtol=0.08, atol=atol_w)
+
+
+def test_probclass_gpu_matches_cpu(dev):
+    """Whole entropy model: GPU gather-conv3d path vs the CPU torch oracle."""
+    import os
+    from dsin_amd import config as cm
+    from dsin_amd.models.probclass import ProbClass
+    here = os.path.dirname(os.path.abspath(__file__))
+    pc_cfg, _ = cm.parse(os.path.join(here, "..", "run_configs", "pc_run_configs"))
+    torch.manual_seed(0)
+    pc = ProbClass(pc_cfg, num_centers=6)
+    q = torch.randn(1, 32, 12, 20)
+    sym = torch.randint(0, 6, (1, 32, 12, 20))
+    bc_cpu = pc.bitcost(q, sym, torch.tensor(0.5))
+    pc_g = pc.to(dev)
+    bc_gpu = pc_g.bitcost(q.to(dev), sym.to(dev), torch.tensor(0.5, device=dev))
+    torch.testing.assert_close(bc_gpu.cpu(), bc_cpu, rtol=0.05, atol=0.05)
