@@ -1,0 +1,86 @@
+"""Fused flat-buffer Adam optimizer.
+
+All parameters of a group are repointed to views of one contiguous fp32
+buffer, gradients to views of a flat gradient buffer (autograd accumulates
+in place into existing .grad tensors), and the whole update is ONE kernel
+launch over the flat buffers per group — versus hundreds of per-tensor
+kernels in a stock optimizer. lr and the step count are device tensors so
+the step replays correctly inside a captured hipGraph while the staircase
+schedule updates lr externally.
+
+Semantics follow tf.train.AdamOptimizer (the reference's optimizer):
+eps is added OUTSIDE the bias-corrected sqrt(v), matching
+lr_t = lr sqrt(1-b2^t)/(1-b1^t); p -= lr_t m/(sqrt(v)+eps).
+
+The flat gradient buffer doubles as the DDP communication buffer: the
+Trainer's reducer can all-reduce it directly.
+"""
+
+from __future__ import annotations
+
+from typing import Iterable, List
+
+import torch
+
+from . import _require_ext
+
+
+class FusedAdam:
+    def __init__(self, params: Iterable[torch.nn.Parameter], lr: float,
+                 betas=(0.9, 0.999), eps: float = 1e-8):
+        self.params: List[torch.nn.Parameter] = [p for p in params]
+        assert self.params, "empty parameter group"
+        device = self.params[0].device
+        assert all(p.dtype == torch.float32 for p in self.params)
+        total = sum(p.numel() for p in self.params)
+
+        self.flat_p = torch.empty(total, device=device)
+        self.flat_g = torch.zeros(total, device=device)
+        self.exp_avg = torch.zeros(total, device=device)
+        self.exp_avg_sq = torch.zeros(total, device=device)
+        ofs = 0
+        self._slices = []
+        for p in self.params:
+            n = p.numel()
+            self.flat_p[ofs:ofs + n].copy_(p.data.reshape(-1))
+            p.data = self.flat_p[ofs:ofs + n].view(p.shape)
+            p.grad = self.flat_g[ofs:ofs + n].view(p.shape)
+            self._slices.append((ofs, n))
+            ofs += n
+
+        self.lr_t = torch.tensor(float(lr), device=device)
+        self.step_t = torch.zeros(1, device=device, dtype=torch.int32)
+        self.betas = tuple(betas)
+        self.eps = float(eps)
+        # torch-optimizer-compatible surface for the LR schedule
+        self.param_groups = [{"lr": self.lr_t, "params": self.params}]
+        self.defaults = {"lr": float(lr)}
+
+    def zero_grad(self, set_to_none: bool = False):
+        # set_to_none would break the flat-buffer invariant; always zero
+        self.flat_g.zero_()
+
+    @torch.no_grad()
+    def step(self):
+        step_fn = _require_ext("adam_step")
+        self.step_t += 1
+        step_fn(self.flat_p, self.flat_g, self.exp_avg, self.exp_avg_sq,
+                self.lr_t, self.step_t, self.betas[0], self.betas[1], self.eps)
+
+    def state_dict(self):
+        return {
+            "flat_p": self.flat_p,
+            "exp_avg": self.exp_avg,
+            "exp_avg_sq": self.exp_avg_sq,
+            "step": self.step_t,
+            "lr": self.lr_t,
+            "betas": self.betas,
+            "eps": self.eps,
+        }
+
+    def load_state_dict(self, sd):
+        self.flat_p.copy_(sd["flat_p"].to(self.flat_p.device))
+        self.exp_avg.copy_(sd["exp_avg"].to(self.exp_avg.device))
+        self.exp_avg_sq.copy_(sd["exp_avg_sq"].to(self.exp_avg_sq.device))
+        self.step_t.copy_(sd["step"].to(self.step_t.device))
+        self.lr_t.copy_(sd["lr"].to(self.lr_t.device))

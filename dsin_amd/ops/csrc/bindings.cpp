@@ -26,6 +26,9 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
 torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t WO, bool mcontig);
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor lr, torch::Tensor step,
+               double b1, double b2, double eps);
 }  // namespace dsin
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -38,4 +41,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_tables", &dsin::conv_tables, "gather-conv offset tables");
   m.def("conv_fwd", &dsin::conv_fwd, "implicit-GEMM gather conv forward");
   m.def("conv_wrw", &dsin::conv_wrw, "implicit-GEMM conv weight gradient");
+  m.def("adam_step", &dsin::adam_step, "fused flat-buffer Adam step");
 }

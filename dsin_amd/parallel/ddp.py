@@ -90,6 +90,8 @@ class GradReducer:
                  comm_dtype: Optional[torch.dtype] = None):
         self.active = is_distributed() and world_size() > 1
         self.params = [p for p in params if p.requires_grad]
+        if not self.params:
+            self.active = False
         self.comm_dtype = comm_dtype
         self._hooks = []
         self._buckets: List[_Bucket] = []

@@ -31,6 +31,11 @@ def _make_optimizer(config, params, device, capturable: bool):
     kind = config.optimizer
     lr = float(config.lr_initial)
     if kind == "ADAM":
+        if device.type == "cuda":
+            from ..ops import hip_available
+            if hip_available():
+                from ..ops.adam import FusedAdam
+                return FusedAdam(params, lr=lr, betas=(0.9, 0.999), eps=1e-8)
         if capturable:
             lr_t = torch.tensor(lr, device=device)
             return torch.optim.Adam(params, lr=lr_t, betas=(0.9, 0.999),
@@ -86,11 +91,22 @@ class Trainer:
         self.sched_pc = LRScheduleT(pc_config, self.opt_pc, itr_ep)
         self.global_step = 0
 
-        self.reducer = GradReducer(ae_params + pc_params,
-                                   bucket_bytes=ddp_bucket_bytes,
-                                   comm_dtype=ddp_comm_dtype)
+        from ..ops.adam import FusedAdam
+        self._fused = isinstance(self.opt_ae, FusedAdam)
+        if self._fused:
+            # flat gradient buffers double as the DDP communication buffers
+            self.reducer = GradReducer([])  # inactive
+        else:
+            self.reducer = GradReducer(ae_params + pc_params,
+                                       bucket_bytes=ddp_bucket_bytes,
+                                       comm_dtype=ddp_comm_dtype)
         if is_distributed():
-            self.reducer.broadcast_params()
+            if self._fused:
+                import torch.distributed as dist
+                for opt in (self.opt_ae, self.opt_pc):
+                    dist.broadcast(opt.flat_p, src=0)
+            else:
+                self.reducer.broadcast_params()
 
         self._graph = None
         self._graph_failed = False
@@ -118,7 +134,14 @@ class Trainer:
         with self._autocast():
             out = self.model.train_losses(x, y)
         out["loss"].backward()
-        self.reducer.finalize()
+        if self._fused and is_distributed():
+            import torch.distributed as dist
+            ws = float(dist.get_world_size())
+            for opt in (self.opt_ae, self.opt_pc):
+                dist.all_reduce(opt.flat_g)
+                opt.flat_g.div_(ws)
+        else:
+            self.reducer.finalize()
         self.opt_ae.step()
         self.opt_pc.step()
         return out["loss"].detach(), out["bpp"].detach()
