@@ -91,15 +91,19 @@ class DSIN(nn.Module):
     def regularization_loss(self) -> torch.Tensor:
         """L2 regularizers: factor * sum(w^2)/2 over encoder+decoder conv
         weights, the centers term, plus PC when enabled (reference
-        src/Distortions_imgcomp.py:129-136, src/quantizer_imgcomp.py:18-24)."""
+        src/Distortions_imgcomp.py:129-136, src/quantizer_imgcomp.py:18-24).
+        Computed as ONE flattened reduction (a per-tensor loop costs ~400
+        kernel launches per step)."""
         dev = self.encoder.quantizer.centers.device
         s = torch.zeros((), device=dev)
         if self.reg_factor:
-            for mod in (self.encoder, self.decoder):
-                for m in mod.modules():
-                    if isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
-                        s = s + 0.5 * (m.weight.float() ** 2).sum()
-            s = s * self.reg_factor
+            if not hasattr(self, "_reg_weights"):
+                self._reg_weights = [
+                    m.weight for mod in (self.encoder, self.decoder)
+                    for m in mod.modules()
+                    if isinstance(m, (nn.Conv2d, nn.ConvTranspose2d))]
+            flat = torch.cat([w.reshape(-1) for w in self._reg_weights])
+            s = self.reg_factor * 0.5 * (flat.float() ** 2).sum()
         s = s + self.encoder.quantizer.regularization_loss()
         s = s + self.probclass.regularization_loss()
         return s

@@ -53,15 +53,42 @@ def _wmat(w1: torch.Tensor) -> torch.Tensor:
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
 
 
-def _padded_buf(x: torch.Tensor, pl: int, pr: int, pt: int, pb: int
-                ) -> torch.Tensor:
-    """Zero-pad NCHW into a buffer with 16 elements of tail slack so the
-    kernel's 16B vector staging may overshoot reads without faulting
-    (garbage lanes are cancelled by weight zero-padding / output masking)."""
+class _PadStuffFn(torch.autograd.Function):
+    """Fused pad + stride-S zero-stuff + bf16 cast in one kernel (pad.hip).
+    Backward is a strided slice of dy (un-pad + un-stuff)."""
+
+    @staticmethod
+    def forward(ctx, x, pt, pb, pl, pr, stride):
+        fn = _require_ext("pad_stuff")
+        ctx.meta = (pt, pl, stride, x.shape, x.dtype)
+        return fn(x.contiguous(), pt, pb, pl, pr, stride)
+
+    @staticmethod
+    def backward(ctx, dy):
+        pt, pl, stride, shape, dtype = ctx.meta
+        B, C, H, W = shape
+        dx = dy[:, :, pt:pt + (H - 1) * stride + 1:stride,
+                pl:pl + (W - 1) * stride + 1:stride]
+        return dx.to(dtype), None, None, None, None, None
+
+
+def _padded_buf(x: torch.Tensor, pl: int, pr: int, pt: int, pb: int,
+                stride: int = 1) -> torch.Tensor:
+    """Zero-pad (and optionally stride-stuff) NCHW into a bf16 buffer with
+    16 elements of tail slack so the conv kernel's 16B vector staging may
+    overshoot reads without faulting (garbage lanes are cancelled by weight
+    zero-padding / output masking)."""
+    if x.is_cuda and hip_available():
+        return _PadStuffFn.apply(x, pt, pb, pl, pr, stride)
+    if stride > 1:
+        b, c, h, w = x.shape
+        z = x.new_zeros(b, c, (h - 1) * stride + 1, (w - 1) * stride + 1)
+        z[..., ::stride, ::stride] = x
+        x = z
     B, C, H, W = x.shape
     Hp, Wp = H + pt + pb, W + pl + pr
     n = C * Hp * Wp
-    store = x.new_zeros(B * n + 16)
+    store = x.new_zeros(B * n + 16, dtype=torch.bfloat16)
     buf = store[:B * n].view(B, C, Hp, Wp)
     buf[:, :, pt:pt + H, pl:pl + W] = x
     return buf
@@ -112,9 +139,8 @@ class _GatherConvFn(torch.autograd.Function):
         dxbuf = None
         if ctx.needs_input_grad[0]:
             with torch.no_grad():
-                z = _zero_stuff(dy, stride)
                 pe_h, pe_w = (kh - 1) * dil, (kw - 1) * dil
-                dybuf = _padded_buf(z, pe_w, pe_w, pe_h, pe_h)
+                dybuf = _padded_buf(dy, pe_w, pe_w, pe_h, pe_h, stride=stride)
                 wrot = (w1.view(Co, Ci, kh, kw).flip(2, 3).permute(1, 0, 2, 3)
                         .reshape(Ci, Co * kh * kw))
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
@@ -149,8 +175,7 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
     Co, _, kh, kw = w.shape
     HO = _out_size(H, kh, stride, padding, dilation)
     WO = _out_size(W, kw, stride, padding, dilation)
-    xbuf = _padded_buf(x.to(torch.bfloat16), padding, padding, padding,
-                       padding)
+    xbuf = _padded_buf(x, padding, padding, padding, padding)
     return _GatherConvFn.apply(xbuf, w.reshape(Co, Ci * kh * kw), bias,
                                stride, dilation, kh, kw, HO, WO, act)
 
@@ -170,10 +195,9 @@ def conv_transpose2d(x: torch.Tensor, w: torch.Tensor,
     _, Co, kh, kw = w.shape
     HO = (H - 1) * stride - 2 * padding + kh + output_padding
     WO = (W - 1) * stride - 2 * padding + kw + output_padding
-    z = _zero_stuff(x.to(torch.bfloat16), stride)
     pl_h, pl_w = kh - 1 - padding, kw - 1 - padding
-    xbuf = _padded_buf(z, pl_w, pl_w + output_padding,
-                       pl_h, pl_h + output_padding)
+    xbuf = _padded_buf(x, pl_w, pl_w + output_padding,
+                       pl_h, pl_h + output_padding, stride=stride)
     w1 = w.flip(2, 3).permute(1, 0, 2, 3).reshape(Co, Ci * kh * kw)
     return _GatherConvFn.apply(xbuf, w1, bias, 1, 1, kh, kw, HO, WO, act)
 

@@ -29,6 +29,8 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor lr, torch::Tensor step,
                double b1, double b2, double eps);
+torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
+                        int64_t pr, int64_t stride);
 }  // namespace dsin
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -42,4 +44,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &dsin::conv_fwd, "implicit-GEMM gather conv forward");
   m.def("conv_wrw", &dsin::conv_wrw, "implicit-GEMM conv weight gradient");
   m.def("adam_step", &dsin::adam_step, "fused flat-buffer Adam step");
+  m.def("pad_stuff", &dsin::pad_stuff, "fused pad/zero-stuff/cast to bf16");
 }

@@ -1,0 +1,79 @@
+#include "hip/hip_runtime.h"
+// Fused pad / zero-stuff / cast kernel: builds the conv input buffer
+// (bf16, zero border, optional stride-S zero-stuffing for transposed convs)
+// in ONE pass from an fp32 or bf16 NCHW tensor — replacing the
+// zeros-fill + cast + interior-copy kernel chain per conv call.
+// The output buffer carries 16 elements of tail slack for the conv staging
+// vector reads (see ops/conv.py).
+
+#include "common_hip.h"
+
+namespace dsin {
+
+template <typename T>
+__global__ void pad_stuff_kernel(const T* __restrict__ x,
+                                 bf16* __restrict__ out,
+                                 int C, int H, int W, int Hp, int Wp,
+                                 int pt, int pl, int stride,
+                                 long long n_img_out, long long n_img_in,
+                                 int B) {
+  long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = (long long)B * n_img_out;
+  long long gstride = (long long)gridDim.x * blockDim.x;
+  for (; idx < total; idx += gstride) {
+    long long b = idx / n_img_out;
+    long long rem = idx % n_img_out;
+    int c = rem / (Hp * Wp);
+    int r2 = rem % (Hp * Wp);
+    int i = r2 / Wp, j = r2 % Wp;
+    float v = 0.f;
+    int ii = i - pt, jj = j - pl;
+    if (ii >= 0 && jj >= 0) {
+      if (stride == 1) {
+        if (ii < H && jj < W)
+          v = (float)x[b * n_img_in + ((long long)c * H + ii) * W + jj];
+      } else {
+        if (ii % stride == 0 && jj % stride == 0) {
+          ii /= stride;
+          jj /= stride;
+          if (ii < H && jj < W)
+            v = (float)x[b * n_img_in + ((long long)c * H + ii) * W + jj];
+        }
+      }
+    }
+    out[idx] = f2b(v);
+  }
+}
+
+torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
+                        int64_t pr, int64_t stride) {
+  CHECK_CUDA_CONTIG(x);
+  TORCH_CHECK(x.dim() == 4, "pad_stuff expects NCHW");
+  const int B = (int)x.size(0), C = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int Hs = (H - 1) * (int)stride + 1, Ws = (W - 1) * (int)stride + 1;
+  const int Hp = Hs + (int)(pt + pb), Wp = Ws + (int)(pl + pr);
+  const long long n_img = (long long)C * Hp * Wp;
+  auto store = torch::empty({(int64_t)B * n_img + 16},
+                            x.options().dtype(torch::kBFloat16));
+  auto out = store.narrow(0, 0, B * n_img).view({B, C, Hp, Wp});
+  long long total = (long long)B * n_img;
+  int grid = (int)std::min<long long>((total + 255) / 256, 8192);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((pad_stuff_kernel<float>), dim3(grid), dim3(256), 0,
+                       stream, x.data_ptr<float>(), (bf16*)out.data_ptr(), C,
+                       H, W, Hp, Wp, (int)pt, (int)pl, (int)stride, n_img,
+                       (long long)C * H * W, B);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((pad_stuff_kernel<bf16>), dim3(grid), dim3(256), 0,
+                       stream, (const bf16*)x.data_ptr(),
+                       (bf16*)out.data_ptr(), C, H, W, Hp, Wp, (int)pt,
+                       (int)pl, (int)stride, n_img, (long long)C * H * W, B);
+  } else {
+    TORCH_CHECK(false, "pad_stuff: fp32 or bf16 only");
+  }
+  return out;
+}
+
+}  // namespace dsin
