@@ -69,14 +69,15 @@ class ConvBNAct(nn.Module):
 
     def forward(self, x):
         from ..ops import conv as dconv
+        from ..ops.bn import batch_norm_act
         if self.transpose:
             y = dconv.conv_transpose2d(x, self.conv.weight, None, self.stride,
                                        self.padding, self.output_padding)
         else:
             y = dconv.conv2d(x, self.conv.weight, None, self.stride,
                              self.padding, 1)
-        y = self.bn(y)
-        return torch.relu(y) if self.relu else y
+        return batch_norm_act(y, self.bn, self.training,
+                              act=1 if self.relu else 0)
 
 
 class ResidualBlock(nn.Module):

@@ -31,6 +31,14 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                double b1, double b2, double eps);
 torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
                         int64_t pr, int64_t stride);
+std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
+                                  torch::Tensor beta, torch::Tensor rmean,
+                                  torch::Tensor rvar, double momentum,
+                                  double eps, bool training, int64_t act);
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
+                                  torch::Tensor out, torch::Tensor mean,
+                                  torch::Tensor rstd, torch::Tensor gamma,
+                                  bool training, int64_t act);
 }  // namespace dsin
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -45,4 +53,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_wrw", &dsin::conv_wrw, "implicit-GEMM conv weight gradient");
   m.def("adam_step", &dsin::adam_step, "fused flat-buffer Adam step");
   m.def("pad_stuff", &dsin::pad_stuff, "fused pad/zero-stuff/cast to bf16");
+  m.def("bn_fwd", &dsin::bn_fwd, "fused batch-norm(+act) forward");
+  m.def("bn_bwd", &dsin::bn_bwd, "fused batch-norm(+act) backward");
 }

@@ -279,3 +279,54 @@ def test_fused_adam_in_trainer(dev):
     loss, bpp = tr.train_step(x, y)
     assert torch.isfinite(loss)
     assert not torch.equal(w_before, model.encoder.h1.conv.weight)
+
+
+# ---------------------------------------------------------------- fused BN
+
+def test_bn_act_matches_torch(dev):
+    from dsin_amd.ops.bn import batch_norm_act
+    torch.manual_seed(0)
+    bn1 = torch.nn.BatchNorm2d(32, eps=1e-5, momentum=0.1).to(dev)
+    bn2 = torch.nn.BatchNorm2d(32, eps=1e-5, momentum=0.1).to(dev)
+    with torch.no_grad():
+        bn2.weight.copy_(bn1.weight)
+        bn2.bias.copy_(bn1.bias)
+        g = torch.rand(32, device=dev) + 0.5
+        bn1.weight.copy_(g); bn2.weight.copy_(g)
+        b = torch.randn(32, device=dev) * 0.3
+        bn1.bias.copy_(b); bn2.bias.copy_(b)
+    x = (torch.randn(2, 32, 20, 24, device=dev) * 2 + 1).to(torch.bfloat16)
+    x1 = x.clone().requires_grad_(True)
+    y1 = batch_norm_act(x1, bn1, training=True, act=1)
+    x2 = x.float().clone().requires_grad_(True)
+    y2 = torch.relu(torch.nn.functional.batch_norm(
+        x2, bn2.running_mean, bn2.running_var, bn2.weight, bn2.bias, True,
+        0.1, 1e-5))
+    torch.testing.assert_close(y1.float(), y2, rtol=0.05, atol=0.05)
+    torch.testing.assert_close(bn1.running_mean, bn2.running_mean,
+                               rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(bn1.running_var, bn2.running_var,
+                               rtol=1e-2, atol=1e-2)
+    gr = torch.randn_like(y2)
+    y1.backward(gr.to(y1.dtype))
+    y2.backward(gr)
+    torch.testing.assert_close(x1.grad.float(), x2.grad, rtol=0.1, atol=0.1)
+    torch.testing.assert_close(bn1.weight.grad, bn2.weight.grad,
+                               rtol=0.05, atol=0.3)
+    torch.testing.assert_close(bn1.bias.grad, bn2.bias.grad,
+                               rtol=0.05, atol=0.3)
+
+
+def test_bn_eval_mode(dev):
+    from dsin_amd.ops.bn import batch_norm_act
+    torch.manual_seed(1)
+    bn = torch.nn.BatchNorm2d(16).to(dev)
+    with torch.no_grad():
+        bn.running_mean.uniform_(-1, 1)
+        bn.running_var.uniform_(0.5, 2)
+    x = torch.randn(1, 16, 8, 12, device=dev).to(torch.bfloat16)
+    y = batch_norm_act(x, bn, training=False, act=0)
+    yr = torch.nn.functional.batch_norm(
+        x.float(), bn.running_mean, bn.running_var, bn.weight, bn.bias,
+        False, 0.1, bn.eps)
+    torch.testing.assert_close(y.float(), yr, rtol=0.05, atol=0.05)
