@@ -1,145 +1,194 @@
 #include "hip/hip_runtime.h"
-// Fused batch-norm (+activation) kernel family.
+// Fused batch-norm (+activation) kernel family, v2.
 //
-// Replaces MIOpen's separate BN-forward / BN-backward / activation kernels
-// for the conv stack. torch.nn.BatchNorm2d semantics: normalize with the
-// BIASED batch variance, running_var updated with the UNBIASED variance,
+// torch.nn.BatchNorm2d semantics: normalize with the BIASED batch variance,
+// running_var updated with the UNBIASED variance,
 // running = (1-momentum)*running + momentum*batch. bf16 data, fp32 stats.
 //
-//  bn_stats:  per-channel mean/rstd (+EMA update, +precomputed scale/shift)
-//  bn_apply:  out = act(scale*y + shift)      [one elementwise pass]
-//  bn_bwd_reduce: per-channel sum(dy_eff), sum(dy_eff * xhat) with the
-//                 activation derivative applied inline from the saved output
-//  bn_bwd_apply:  dx = gamma*rstd*(dy_eff - s1/n - xhat * s2/n); also emits
-//                 dgamma = s2, dbeta = s1
+// Layout-aware: NCHW is channel-major, so every kernel runs a 2D grid
+// (spatial chunks x B*C) — the channel index is one divide per BLOCK and
+// all loads are 16B-vectorized.
+//
+//  bn_stats_part -> bn_finalize: per-channel mean/rstd (+EMA, scale/shift)
+//  bn_apply:       out = act(scale*y + shift)
+//  bn_bwd_part:    per-channel sum(dy_eff), sum(dy_eff*xhat), act' inline
+//  bn_bwd_apply:   dx = gamma*rstd*(dy_eff - s1/n - xhat*s2/n)
 
 #include "common_hip.h"
 
 namespace dsin {
 
 using bnbf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(8))) unsigned short bn_u16x8;
 
-__global__ void bn_stats_kernel(const bnbf16* __restrict__ y,
-                                float* __restrict__ mean,
-                                float* __restrict__ rstd,
-                                float* __restrict__ scale,
-                                float* __restrict__ shift,
-                                const float* __restrict__ gamma,
-                                const float* __restrict__ beta,
-                                float* __restrict__ rmean,
-                                float* __restrict__ rvar,
-                                int C, long long HW, int B, long long cstride,
-                                float momentum, float eps, int training) {
-  const int c = blockIdx.x;
-  float s = 0.f, s2 = 0.f;
-  if (training) {
-    for (int b = 0; b < B; ++b) {
-      const bnbf16* p = y + b * cstride * C + c * cstride;
-      for (long long i = threadIdx.x; i < HW; i += blockDim.x) {
-        float v = __bfloat162float(p[i]);
-        s += v;
-        s2 += v * v;
-      }
-    }
-    s = wave_reduce_sum(s);
-    s2 = wave_reduce_sum(s2);
-    __shared__ float red[2][4];
-    const int wid = threadIdx.x >> 6;
-    if ((threadIdx.x & 63) == 0) {
-      red[0][wid] = s;
-      red[1][wid] = s2;
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      float ts = 0.f, ts2 = 0.f;
-      for (int w = 0; w < (int)(blockDim.x >> 6); ++w) {
-        ts += red[0][w];
-        ts2 += red[1][w];
-      }
-      const float n = (float)(HW * B);
-      const float m = ts / n;
-      float var = ts2 / n - m * m;
-      var = fmaxf(var, 0.f);
-      const float rs = rsqrtf(var + eps);
-      mean[c] = m;
-      rstd[c] = rs;
-      // torch: running_var uses the unbiased estimate
-      const float unbias = (n > 1.f) ? var * n / (n - 1.f) : var;
-      rmean[c] = (1.f - momentum) * rmean[c] + momentum * m;
-      rvar[c] = (1.f - momentum) * rvar[c] + momentum * unbias;
-      const float sc = gamma[c] * rs;
-      scale[c] = sc;
-      shift[c] = beta[c] - m * sc;
-    }
-  } else if (threadIdx.x == 0) {
-    const float m = rmean[c];
-    const float rs = rsqrtf(rvar[c] + eps);
-    mean[c] = m;
-    rstd[c] = rs;
-    const float sc = gamma[c] * rs;
-    scale[c] = sc;
-    shift[c] = beta[c] - m * sc;
+__device__ __forceinline__ float bnb2f(unsigned short u) {
+  bnbf16 v = *reinterpret_cast<bnbf16*>(&u);
+  return __bfloat162float(v);
+}
+
+__device__ __forceinline__ void block_reduce2(float& a, float& b) {
+  a = wave_reduce_sum(a);
+  b = wave_reduce_sum(b);
+  __shared__ float red[2][4];
+  const int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) {
+    red[0][wid] = a;
+    red[1][wid] = b;
   }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < (int)(blockDim.x >> 6); ++w) {
+      a += red[0][w];
+      b += red[1][w];
+    }
+  }
+}
+
+__global__ void bn_stats_part_kernel(const bnbf16* __restrict__ y,
+                                     float* __restrict__ psum,   // (C,)
+                                     float* __restrict__ psum2,  // (C,)
+                                     int C, long long HW, int B, int S) {
+  const int c = blockIdx.y;
+  const int sidx = blockIdx.x;
+  float s = 0.f, s2 = 0.f;
+  for (int b = 0; b < B; ++b) {
+    const bnbf16* p = y + ((long long)b * C + c) * HW;
+    long long i0 = (long long)(sidx * (int)blockDim.x + threadIdx.x) * 8;
+    long long stride = (long long)S * blockDim.x * 8;
+    for (long long i = i0; i + 7 < HW; i += stride) {
+      const bn_u16x8 v = *reinterpret_cast<const bn_u16x8*>(&p[i]);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float f = bnb2f(v[k]);
+        s += f;
+        s2 += f * f;
+      }
+    }
+    if (sidx == 0 && threadIdx.x < (HW & 7)) {  // tail
+      float f = __bfloat162float(p[(HW & ~7LL) + threadIdx.x]);
+      s += f;
+      s2 += f * f;
+    }
+  }
+  block_reduce2(s, s2);
+  if (threadIdx.x == 0) {
+    atomicAdd(&psum[c], s);
+    atomicAdd(&psum2[c], s2);
+  }
+}
+
+__global__ void bn_finalize_kernel(const float* __restrict__ psum,
+                                   const float* __restrict__ psum2,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ rstd,
+                                   float* __restrict__ scale,
+                                   float* __restrict__ shift,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ beta,
+                                   float* __restrict__ rmean,
+                                   float* __restrict__ rvar,
+                                   int C, float n, float momentum, float eps,
+                                   int training) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m, rs;
+  if (training) {
+    m = psum[c] / n;
+    float var = fmaxf(psum2[c] / n - m * m, 0.f);
+    rs = rsqrtf(var + eps);
+    const float unbias = (n > 1.f) ? var * n / (n - 1.f) : var;
+    rmean[c] = (1.f - momentum) * rmean[c] + momentum * m;
+    rvar[c] = (1.f - momentum) * rvar[c] + momentum * unbias;
+  } else {
+    m = rmean[c];
+    rs = rsqrtf(rvar[c] + eps);
+  }
+  mean[c] = m;
+  rstd[c] = rs;
+  const float sc = gamma[c] * rs;
+  scale[c] = sc;
+  shift[c] = beta[c] - m * sc;
 }
 
 __global__ void bn_apply_kernel(const bnbf16* __restrict__ y,
                                 bnbf16* __restrict__ out,
                                 const float* __restrict__ scale,
                                 const float* __restrict__ shift,
-                                int C, long long HW, long long total,
-                                int act) {
-  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  long long stride = (long long)gridDim.x * blockDim.x;
-  for (; i < total; i += stride) {
-    const int c = (int)((i / HW) % C);
-    float v = __bfloat162float(y[i]) * scale[c] + shift[c];
-    if (act == 1) v = fmaxf(v, 0.f);
-    else if (act == 2) v = fmaxf(v, 0.2f * v);
-    out[i] = __float2bfloat16(v);
+                                int C, long long HW, int act) {
+  const int bc = blockIdx.y;
+  const int c = bc % C;
+  const float sc = scale[c], sh = shift[c];
+  const bnbf16* p = y + (long long)bc * HW;
+  bnbf16* o = out + (long long)bc * HW;
+  long long i0 = (long long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  long long stride = (long long)gridDim.x * blockDim.x * 8;
+  for (long long i = i0; i + 7 < HW; i += stride) {
+    const bn_u16x8 v = *reinterpret_cast<const bn_u16x8*>(&p[i]);
+    bn_u16x8 r;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float f = bnb2f(v[k]) * sc + sh;
+      if (act == 1) f = fmaxf(f, 0.f);
+      else if (act == 2) f = fmaxf(f, 0.2f * f);
+      bnbf16 h = __float2bfloat16(f);
+      r[k] = *reinterpret_cast<unsigned short*>(&h);
+    }
+    *reinterpret_cast<bn_u16x8*>(&o[i]) = r;
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (HW & 7)) {
+    long long j = (HW & ~7LL) + threadIdx.x;
+    float f = __bfloat162float(p[j]) * sc + sh;
+    if (act == 1) f = fmaxf(f, 0.f);
+    else if (act == 2) f = fmaxf(f, 0.2f * f);
+    o[j] = __float2bfloat16(f);
   }
 }
 
-__global__ void bn_bwd_reduce_kernel(const bnbf16* __restrict__ dy,
-                                     const bnbf16* __restrict__ y,
-                                     const bnbf16* __restrict__ out,
-                                     const float* __restrict__ mean,
-                                     const float* __restrict__ rstd,
-                                     float* __restrict__ s1,  // (C,) sum dy
-                                     float* __restrict__ s2,  // (C,) sum dy*xh
-                                     int C, long long HW, int B,
-                                     long long cstride, int act) {
-  const int c = blockIdx.x;
+__device__ __forceinline__ float bn_actp(float g, float ov, int act) {
+  if (act == 1) return ov > 0.f ? g : 0.f;
+  if (act == 2) return ov > 0.f ? g : 0.2f * g;
+  return g;
+}
+
+__global__ void bn_bwd_part_kernel(const bnbf16* __restrict__ dy,
+                                   const bnbf16* __restrict__ y,
+                                   const bnbf16* __restrict__ out,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ rstd,
+                                   float* __restrict__ s1,
+                                   float* __restrict__ s2,
+                                   int C, long long HW, int B, int S,
+                                   int act) {
+  const int c = blockIdx.y;
+  const int sidx = blockIdx.x;
   const float m = mean[c], rs = rstd[c];
   float a = 0.f, b2 = 0.f;
   for (int b = 0; b < B; ++b) {
-    const long long base = b * cstride * C + c * cstride;
-    for (long long i = threadIdx.x; i < HW; i += blockDim.x) {
-      float g = __bfloat162float(dy[base + i]);
-      if (act == 1) g = (__bfloat162float(out[base + i]) > 0.f) ? g : 0.f;
-      else if (act == 2)
-        g = (__bfloat162float(out[base + i]) > 0.f) ? g : 0.2f * g;
-      const float xh = (__bfloat162float(y[base + i]) - m) * rs;
+    const long long base = ((long long)b * C + c) * HW;
+    long long i0 = (long long)(sidx * (int)blockDim.x + threadIdx.x) * 8;
+    long long stride = (long long)S * blockDim.x * 8;
+    for (long long i = i0; i + 7 < HW; i += stride) {
+      const bn_u16x8 gv = *reinterpret_cast<const bn_u16x8*>(&dy[base + i]);
+      const bn_u16x8 yv = *reinterpret_cast<const bn_u16x8*>(&y[base + i]);
+      const bn_u16x8 ov = *reinterpret_cast<const bn_u16x8*>(&out[base + i]);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = bn_actp(bnb2f(gv[k]), bnb2f(ov[k]), act);
+        a += g;
+        b2 += g * (bnb2f(yv[k]) - m) * rs;
+      }
+    }
+    if (sidx == 0 && threadIdx.x < (HW & 7)) {
+      long long j = base + (HW & ~7LL) + threadIdx.x;
+      float g = bn_actp(__bfloat162float(dy[j]), __bfloat162float(out[j]), act);
       a += g;
-      b2 += g * xh;
+      b2 += g * (__bfloat162float(y[j]) - m) * rs;
     }
   }
-  a = wave_reduce_sum(a);
-  b2 = wave_reduce_sum(b2);
-  __shared__ float red[2][4];
-  const int wid = threadIdx.x >> 6;
-  if ((threadIdx.x & 63) == 0) {
-    red[0][wid] = a;
-    red[1][wid] = b2;
-  }
-  __syncthreads();
+  block_reduce2(a, b2);
   if (threadIdx.x == 0) {
-    float ta = 0.f, tb = 0.f;
-    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) {
-      ta += red[0][w];
-      tb += red[1][w];
-    }
-    s1[c] = ta;
-    s2[c] = tb;
+    atomicAdd(&s1[c], a);
+    atomicAdd(&s2[c], b2);
   }
 }
 
@@ -152,27 +201,54 @@ __global__ void bn_bwd_apply_kernel(const bnbf16* __restrict__ dy,
                                     const float* __restrict__ gamma,
                                     const float* __restrict__ s1,
                                     const float* __restrict__ s2,
-                                    int C, long long HW, long long total,
-                                    float invn, int act, int training) {
-  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  long long stride = (long long)gridDim.x * blockDim.x;
-  for (; i < total; i += stride) {
-    const int c = (int)((i / HW) % C);
-    float g = __bfloat162float(dy[i]);
-    if (act == 1) g = (__bfloat162float(out[i]) > 0.f) ? g : 0.f;
-    else if (act == 2) g = (__bfloat162float(out[i]) > 0.f) ? g : 0.2f * g;
+                                    int C, long long HW, float invn, int act,
+                                    int training) {
+  const int bc = blockIdx.y;
+  const int c = bc % C;
+  const float m = mean[c], rs = rstd[c], gm = gamma[c];
+  const float t1 = s1[c] * invn, t2 = s2[c] * invn;
+  const long long base = (long long)bc * HW;
+  long long i0 = (long long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  long long stride = (long long)gridDim.x * blockDim.x * 8;
+  for (long long i = i0; i + 7 < HW; i += stride) {
+    const bn_u16x8 gv = *reinterpret_cast<const bn_u16x8*>(&dy[base + i]);
+    const bn_u16x8 ov = *reinterpret_cast<const bn_u16x8*>(&out[base + i]);
+    const bn_u16x8 yv = *reinterpret_cast<const bn_u16x8*>(&y[base + i]);
+    bn_u16x8 r;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float g = bn_actp(bnb2f(gv[k]), bnb2f(ov[k]), act);
+      float v;
+      if (training) {
+        const float xh = (bnb2f(yv[k]) - m) * rs;
+        v = gm * rs * (g - t1 - xh * t2);
+      } else {
+        v = gm * rs * g;
+      }
+      bnbf16 h = __float2bfloat16(v);
+      r[k] = *reinterpret_cast<unsigned short*>(&h);
+    }
+    *reinterpret_cast<bn_u16x8*>(&dx[base + i]) = r;
+  }
+  if (blockIdx.x == 0 && threadIdx.x < (HW & 7)) {
+    long long j = base + (HW & ~7LL) + threadIdx.x;
+    float g = bn_actp(__bfloat162float(dy[j]), __bfloat162float(out[j]), act);
     float v;
     if (training) {
-      const float xh = (__bfloat162float(y[i]) - mean[c]) * rstd[c];
-      v = gamma[c] * rstd[c] * (g - s1[c] * invn - xh * (s2[c] * invn));
+      const float xh = (__bfloat162float(y[j]) - m) * rs;
+      v = gm * rs * (g - t1 - xh * t2);
     } else {
-      v = gamma[c] * rstd[c] * g;  // eval: stats are constants
+      v = gm * rs * g;
     }
-    dx[i] = __float2bfloat16(v);
+    dx[j] = __float2bfloat16(v);
   }
 }
 
 // --------------------------------------------------------------- host
+
+static int _spatial_chunks(long long HW) {
+  return (int)std::min<long long>(std::max<long long>(HW / (256 * 8), 1), 32);
+}
 
 std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
                                   torch::Tensor beta, torch::Tensor rmean,
@@ -188,20 +264,36 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
   auto scale = torch::empty({C}, optsF);
   auto shift = torch::empty({C}, optsF);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(bn_stats_kernel, dim3(C), dim3(256), 0, stream,
-                     (const bnbf16*)y.data_ptr(), mean.data_ptr<float>(),
-                     rstd.data_ptr<float>(), scale.data_ptr<float>(),
-                     shift.data_ptr<float>(), gamma.data_ptr<float>(),
-                     beta.data_ptr<float>(), rmean.data_ptr<float>(),
-                     rvar.data_ptr<float>(), C, HW, B, HW,
-                     (float)momentum, (float)eps, training ? 1 : 0);
+  if (training) {
+    auto psum = torch::zeros({C}, optsF);
+    auto psum2 = torch::zeros({C}, optsF);
+    const int S = _spatial_chunks(HW);
+    hipLaunchKernelGGL(bn_stats_part_kernel, dim3(S, C), dim3(256), 0, stream,
+                       (const bnbf16*)y.data_ptr(), psum.data_ptr<float>(),
+                       psum2.data_ptr<float>(), C, HW, B, S);
+    hipLaunchKernelGGL(bn_finalize_kernel, grid1d(C, 256), dim3(256), 0,
+                       stream, psum.data_ptr<float>(), psum2.data_ptr<float>(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       rmean.data_ptr<float>(), rvar.data_ptr<float>(), C,
+                       (float)((long long)B * HW), (float)momentum, (float)eps,
+                       1);
+  } else {
+    hipLaunchKernelGGL(bn_finalize_kernel, grid1d(C, 256), dim3(256), 0,
+                       stream, nullptr, nullptr, mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), scale.data_ptr<float>(),
+                       shift.data_ptr<float>(), gamma.data_ptr<float>(),
+                       beta.data_ptr<float>(), rmean.data_ptr<float>(),
+                       rvar.data_ptr<float>(), C, (float)((long long)B * HW),
+                       (float)momentum, (float)eps, 0);
+  }
   auto out = torch::empty_like(y);
-  long long total = (long long)B * C * HW;
-  int grid = (int)std::min<long long>((total + 255) / 256, 8192);
-  hipLaunchKernelGGL(bn_apply_kernel, dim3(grid), dim3(256), 0, stream,
+  hipLaunchKernelGGL(bn_apply_kernel,
+                     dim3(_spatial_chunks(HW), B * C), dim3(256), 0, stream,
                      (const bnbf16*)y.data_ptr(), (bnbf16*)out.data_ptr(),
                      scale.data_ptr<float>(), shift.data_ptr<float>(), C, HW,
-                     total, (int)act);
+                     (int)act);
   return {out, mean, rstd};
 }
 
@@ -214,23 +306,22 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   const int B = (int)y.size(0), C = (int)y.size(1);
   const long long HW = (long long)y.size(2) * y.size(3);
   auto optsF = y.options().dtype(torch::kFloat32);
-  auto s1 = torch::empty({C}, optsF);
-  auto s2 = torch::empty({C}, optsF);
+  auto s1 = torch::zeros({C}, optsF);
+  auto s2 = torch::zeros({C}, optsF);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C), dim3(256), 0, stream,
+  const int S = _spatial_chunks(HW);
+  hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(S, C), dim3(256), 0, stream,
                      (const bnbf16*)dy.data_ptr(), (const bnbf16*)y.data_ptr(),
                      (const bnbf16*)out.data_ptr(), mean.data_ptr<float>(),
                      rstd.data_ptr<float>(), s1.data_ptr<float>(),
-                     s2.data_ptr<float>(), C, HW, B, HW, (int)act);
+                     s2.data_ptr<float>(), C, HW, B, S, (int)act);
   auto dx = torch::empty_like(y);
-  long long total = (long long)B * C * HW;
-  int grid = (int)std::min<long long>((total + 255) / 256, 8192);
-  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(grid), dim3(256), 0, stream,
-                     (const bnbf16*)dy.data_ptr(), (const bnbf16*)y.data_ptr(),
-                     (const bnbf16*)out.data_ptr(), (bnbf16*)dx.data_ptr(),
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     gamma.data_ptr<float>(), s1.data_ptr<float>(),
-                     s2.data_ptr<float>(), C, HW, total,
+  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(_spatial_chunks(HW), B * C),
+                     dim3(256), 0, stream, (const bnbf16*)dy.data_ptr(),
+                     (const bnbf16*)y.data_ptr(), (const bnbf16*)out.data_ptr(),
+                     (bnbf16*)dx.data_ptr(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                     s1.data_ptr<float>(), s2.data_ptr<float>(), C, HW,
                      1.f / (float)(B * HW), (int)act, training ? 1 : 0);
   // dgamma = s2, dbeta = s1
   return {dx, s2, s1};
