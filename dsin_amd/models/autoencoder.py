@@ -67,7 +67,7 @@ class ConvBNAct(nn.Module):
         self.bn = _bn(cout)
         self.relu = relu
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         from ..ops import conv as dconv
         from ..ops.bn import batch_norm_act
         if self.transpose:
@@ -77,7 +77,7 @@ class ConvBNAct(nn.Module):
             y = dconv.conv2d(x, self.conv.weight, None, self.stride,
                              self.padding, 1)
         return batch_norm_act(y, self.bn, self.training,
-                              act=1 if self.relu else 0)
+                              act=1 if self.relu else 0, residual=residual)
 
 
 class ResidualBlock(nn.Module):
@@ -92,7 +92,8 @@ class ResidualBlock(nn.Module):
         self.c2 = ConvBNAct(ch, ch, 3, relu=False)
 
     def forward(self, x):
-        return self.c2(self.c1(x)) + x
+        # skip-add fused into the second conv's BN apply pass
+        return self.c2(self.c1(x), residual=x)
 
 
 class ResidualStack(nn.Module):

@@ -34,7 +34,8 @@ torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
 std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
                                   torch::Tensor beta, torch::Tensor rmean,
                                   torch::Tensor rvar, double momentum,
-                                  double eps, bool training, int64_t act);
+                                  double eps, bool training, int64_t act,
+                                  c10::optional<torch::Tensor> residual);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                   torch::Tensor out, torch::Tensor mean,
                                   torch::Tensor rstd, torch::Tensor gamma,

@@ -111,6 +111,7 @@ __global__ void bn_finalize_kernel(const float* __restrict__ psum,
 }
 
 __global__ void bn_apply_kernel(const bnbf16* __restrict__ y,
+                                const bnbf16* __restrict__ res,  // or null
                                 bnbf16* __restrict__ out,
                                 const float* __restrict__ scale,
                                 const float* __restrict__ shift,
@@ -119,15 +120,19 @@ __global__ void bn_apply_kernel(const bnbf16* __restrict__ y,
   const int c = bc % C;
   const float sc = scale[c], sh = shift[c];
   const bnbf16* p = y + (long long)bc * HW;
+  const bnbf16* q = res ? res + (long long)bc * HW : nullptr;
   bnbf16* o = out + (long long)bc * HW;
   long long i0 = (long long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
   long long stride = (long long)gridDim.x * blockDim.x * 8;
   for (long long i = i0; i + 7 < HW; i += stride) {
     const bn_u16x8 v = *reinterpret_cast<const bn_u16x8*>(&p[i]);
+    bn_u16x8 rv;
+    if (q) rv = *reinterpret_cast<const bn_u16x8*>(&q[i]);
     bn_u16x8 r;
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
       float f = bnb2f(v[k]) * sc + sh;
+      if (q) f += bnb2f(rv[k]);
       if (act == 1) f = fmaxf(f, 0.f);
       else if (act == 2) f = fmaxf(f, 0.2f * f);
       bnbf16 h = __float2bfloat16(f);
@@ -138,6 +143,7 @@ __global__ void bn_apply_kernel(const bnbf16* __restrict__ y,
   if (blockIdx.x == 0 && threadIdx.x < (HW & 7)) {
     long long j = (HW & ~7LL) + threadIdx.x;
     float f = __bfloat162float(p[j]) * sc + sh;
+    if (q) f += __bfloat162float(q[j]);
     if (act == 1) f = fmaxf(f, 0.f);
     else if (act == 2) f = fmaxf(f, 0.2f * f);
     o[j] = __float2bfloat16(f);
@@ -253,7 +259,8 @@ static int _spatial_chunks(long long HW) {
 std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
                                   torch::Tensor beta, torch::Tensor rmean,
                                   torch::Tensor rvar, double momentum,
-                                  double eps, bool training, int64_t act) {
+                                  double eps, bool training, int64_t act,
+                                  c10::optional<torch::Tensor> residual) {
   CHECK_CUDA_CONTIG(y);
   TORCH_CHECK(y.scalar_type() == torch::kBFloat16, "bn: y must be bf16");
   const int B = (int)y.size(0), C = (int)y.size(1);
@@ -289,9 +296,14 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
                        (float)momentum, (float)eps, 0);
   }
   auto out = torch::empty_like(y);
+  const bnbf16* resp = nullptr;
+  if (residual.has_value()) {
+    CHECK_CUDA_CONTIG(residual.value());
+    resp = (const bnbf16*)residual->data_ptr();
+  }
   hipLaunchKernelGGL(bn_apply_kernel,
                      dim3(_spatial_chunks(HW), B * C), dim3(256), 0, stream,
-                     (const bnbf16*)y.data_ptr(), (bnbf16*)out.data_ptr(),
+                     (const bnbf16*)y.data_ptr(), resp, (bnbf16*)out.data_ptr(),
                      scale.data_ptr<float>(), shift.data_ptr<float>(), C, HW,
                      (int)act);
   return {out, mean, rstd};
