@@ -1,0 +1,83 @@
+"""Range coder + entropy-model codec tests (CPU)."""
+
+import numpy as np
+import torch
+import pytest
+
+from dsin_amd.coding import (PredictionNetwork, ProbclassTesting,
+                             decode_symbols, decode_with_freqs,
+                             encode_symbols, encode_with_freqs)
+from dsin_amd.models.probclass import ProbClass
+
+
+def test_range_coder_roundtrip_uniform():
+    rng = np.random.default_rng(0)
+    n = 2000
+    syms = rng.integers(0, 6, n)
+    freqs = np.ones(6, dtype=np.int64) * 100
+    data = encode_with_freqs(syms, (freqs for _ in range(n)))
+    out = decode_with_freqs(data, n, lambda i, prev: freqs)
+    assert list(syms) == out
+
+
+def test_range_coder_roundtrip_skewed():
+    rng = np.random.default_rng(1)
+    n = 3000
+    freqs = np.array([60000, 3000, 1500, 800, 400, 300], dtype=np.int64)
+    p = freqs / freqs.sum()
+    syms = rng.choice(6, size=n, p=p)
+    data = encode_with_freqs(syms, (freqs for _ in range(n)))
+    out = decode_with_freqs(data, n, lambda i, prev: freqs)
+    assert list(syms) == out
+    # compression close to entropy
+    ent_bits = -np.sum(np.log2(p[syms]))
+    assert len(data) * 8 < ent_bits * 1.02 + 64
+
+
+def _make_pc(pc_config):
+    torch.manual_seed(0)
+    return ProbClass(pc_config, num_centers=6)
+
+
+def test_probclass_testing_matches_bitcost(pc_config):
+    pc = _make_pc(pc_config)
+    centers = torch.linspace(-2, 2, 6)
+    symbols = torch.randint(0, 6, (4, 6, 8))
+    pt = ProbclassTesting(pc, centers)
+    total = pt.total_bit_cost(symbols)
+    q = centers[symbols.unsqueeze(0)]
+    bc = pc.bitcost(q, symbols.unsqueeze(0), centers[0])
+    assert abs(total - float(bc.sum())) < 1e-3
+
+
+def test_prediction_network_shapes(pc_config):
+    pc = _make_pc(pc_config)
+    centers = torch.linspace(-2, 2, 6)
+    pred = PredictionNetwork(pc, centers)
+    assert pred.context_shape == (5, 9, 9)
+    ctx = torch.randint(0, 6, pred.context_shape)
+    f = pred.freqs(ctx)
+    assert f.shape == (6,) and (f >= 1).all()
+
+
+def test_codec_roundtrip_exact(pc_config):
+    """Bit-exact autoregressive encode/decode through the entropy model."""
+    pc = _make_pc(pc_config)
+    centers = torch.linspace(-2, 2, 6)
+    torch.manual_seed(2)
+    symbols = torch.randint(0, 6, (3, 5, 6))
+    data = encode_symbols(pc, centers, symbols, exact=True)
+    out = decode_symbols(pc, centers, data, (3, 5, 6))
+    assert torch.equal(out.cpu(), symbols)
+
+
+def test_codec_bits_close_to_bitcost(pc_config):
+    pc = _make_pc(pc_config)
+    centers = torch.linspace(-2, 2, 6)
+    torch.manual_seed(3)
+    symbols = torch.randint(0, 6, (4, 8, 10))
+    data = encode_symbols(pc, centers, symbols, exact=False)
+    est_bits = ProbclassTesting(pc, centers).total_bit_cost(symbols)
+    actual_bits = len(data) * 8
+    # coder overhead: freq quantization + 4 flush bytes
+    assert actual_bits < est_bits * 1.05 + 64, (actual_bits, est_bits)
