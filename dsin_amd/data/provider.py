@@ -178,3 +178,49 @@ class SyntheticStereo:
         x = x.clamp(0, 255).to(self.device)
         y = y.clamp(0, 255).to(self.device)
         return x, y
+
+
+class SyntheticDataset:
+    """Dataset-compatible synthetic source (train/val/test views) for runs
+    without KITTI on disk (`root_data = synthetic` in the ae config). Val and
+    test iterate fixed seeds so metrics are reproducible."""
+
+    def __init__(self, config, n_train: int = 1576, n_val: int = 16,
+                 n_test: int = 16):
+        crop_h, crop_w = config.crop_size
+        self.batch_size = config.batch_size if config.AE_only else 1
+        self.train_pairs = [("synthetic", str(i)) for i in range(n_train)]
+        self.val_pairs = [("synthetic", str(i)) for i in range(n_val)]
+        self.test_pairs = [("synthetic", str(i)) for i in range(n_test)]
+        self._train = SyntheticStereo(crop_h, crop_w, self.batch_size, seed=1)
+        self._val = SyntheticStereo(crop_h, crop_w, self.batch_size, seed=2)
+        self._test = SyntheticStereo(crop_h, crop_w, self.batch_size, seed=3)
+        self._n_val, self._n_test = n_val, n_test
+
+    def get_data_size(self):
+        return self.val_pairs, self.test_pairs
+
+    def get_data_for_train(self):
+        return self._train.next_batch()
+
+    def get_data_for_val(self):
+        if self._val._step >= self._n_val:
+            self._val._step = 0
+        return self._val.next_batch()
+
+    def get_data_for_test(self):
+        if self._test._step >= self._n_test:
+            self._test._step = 0
+        return self._test.next_batch()
+
+
+def make_dataset(config, data_paths_dir: str, seed: int = 0):
+    """Dataset factory: path-pair lists when present, synthetic otherwise."""
+    if str(config.root_data) == "synthetic":
+        return SyntheticDataset(config)
+    train_list = os.path.join(data_paths_dir, config.file_path_train)
+    if not os.path.exists(train_list):
+        import warnings
+        warnings.warn(f"{train_list} not found - using synthetic data")
+        return SyntheticDataset(config)
+    return Dataset(config, data_paths_dir, seed=seed)

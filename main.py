@@ -16,7 +16,7 @@ import numpy as np
 import torch
 
 from dsin_amd import config as config_mod
-from dsin_amd.data import Dataset
+from dsin_amd.data import make_dataset
 from dsin_amd.models import DSIN
 from dsin_amd.parallel import init_distributed, is_distributed, rank
 from dsin_amd.training import Trainer, checkpoint
@@ -32,7 +32,7 @@ def get_validate_every(iteration, total_iterations, validate_every, p1, p2):
     if iteration > 3 * (total_iterations // 4) and not p2:
         validate_every //= 2
         p2 = True
-    return validate_every, p1, p2
+    return max(validate_every, 1), p1, p2
 
 
 def main(argv=None):
@@ -60,7 +60,7 @@ def main(argv=None):
     os.makedirs(root_weights, exist_ok=True)
 
     model = DSIN(ae_config, pc_config).to(device)
-    data = Dataset(ae_config, data_dir, seed=rank())
+    data = make_dataset(ae_config, data_dir, seed=rank())
     num_train = len(data.train_pairs)
     trainer = Trainer(model, ae_config, pc_config, num_train, device=device,
                      autocast_bf16=args.bf16)
