@@ -27,13 +27,25 @@ def _gauss_kernel1d(sigma: float, size: int, device, dtype) -> torch.Tensor:
     return g / g.abs().sum()
 
 
+def _conv_valid(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """VALID conv used by the blur passes. On GPU the per-channel grouped
+    conv is flattened to batch B*C with Ci=Co=1 and routed through the
+    custom gather-GEMM kernel (keeps MIOpen out of the ms_ssim loss path)."""
+    if x.is_cuda:
+        from ..ops import hip_available
+        from ..ops import conv as dconv
+        if hip_available():
+            b, c, h, wd = x.shape
+            y = dconv.conv2d(x.reshape(b * c, 1, h, wd), w, None, 1, 0, 1)
+            return y.reshape(b, c, y.shape[2], y.shape[3]).float()
+    c = x.shape[1]
+    return F.conv2d(x, w.expand(c, 1, w.shape[2], w.shape[3]), groups=c)
+
+
 def _blur_sep(x: torch.Tensor, k1d: torch.Tensor) -> torch.Tensor:
     """Per-channel separable VALID conv, rows then cols (reference :31-42)."""
-    c = x.shape[1]
-    kh = k1d.view(1, 1, 1, -1).expand(c, 1, 1, -1)
-    kv = k1d.view(1, 1, -1, 1).expand(c, 1, -1, 1)
-    x = F.conv2d(x, kh, groups=c)
-    return F.conv2d(x, kv, groups=c)
+    x = _conv_valid(x, k1d.view(1, 1, 1, -1))
+    return _conv_valid(x, k1d.view(1, 1, -1, 1))
 
 
 def _ssim_scale(img1: torch.Tensor, img2: torch.Tensor, max_val: float,
@@ -60,12 +72,9 @@ def _ssim_scale(img1: torch.Tensor, img2: torch.Tensor, max_val: float,
 def _downsample2(x: torch.Tensor) -> torch.Tensor:
     """2-tap box filter, REFLECT pad (0 front, 1 back), separable, then ::2
     (reference kernel_blur at :46-64 with pad_w1=0, pad_w2=1)."""
-    c = x.shape[1]
     x = F.pad(x, (0, 1, 0, 1), mode="reflect")
-    kh = x.new_full((c, 1, 1, 2), 0.5)
-    kv = x.new_full((c, 1, 2, 1), 0.5)
-    x = F.conv2d(x, kh, groups=c)
-    x = F.conv2d(x, kv, groups=c)
+    x = _conv_valid(x, x.new_full((1, 1, 1, 2), 0.5))
+    x = _conv_valid(x, x.new_full((1, 1, 2, 1), 0.5))
     return x[:, :, ::2, ::2]
 
 

@@ -74,5 +74,11 @@ def load(model, optimizers, load_path: str, ae_config) -> int:
         ts = blob["training-step"]
         step = int(ts.get("global_step", 0))
         for opt, sd in zip(optimizers, ts.get("optimizers", [])):
-            opt.load_state_dict(sd)
+            try:
+                opt.load_state_dict(sd)
+            except (KeyError, ValueError, RuntimeError) as e:
+                import warnings
+                warnings.warn(
+                    "optimizer state not restored (checkpoint written by a "
+                    f"different optimizer implementation): {e}")
     return step
