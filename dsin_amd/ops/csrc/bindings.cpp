@@ -16,6 +16,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ncc_search(
     torch::Tensor x_dec, torch::Tensor y_dec, torch::Tensor y_orig, int64_t ph,
     int64_t pw, bool use_mask);
 torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B);
+torch::Tensor mfma32_selftest(torch::Tensor A, torch::Tensor B);
 std::tuple<torch::Tensor, torch::Tensor> conv_tables(
     int64_t M, int64_t K, int64_t WO, int64_t stride, int64_t dil, int64_t Wp,
     int64_t HpWp, int64_t kh, int64_t kw, torch::Device device);
@@ -49,6 +50,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bitcost_ce_bwd", &dsin::bitcost_ce_bwd, "bitcost cross-entropy bwd");
   m.def("ncc_search", &dsin::ncc_search, "streaming NCC side-info search");
   m.def("mfma_selftest", &dsin::mfma_selftest, "MFMA 16x16x32 layout check");
+  m.def("mfma32_selftest", &dsin::mfma32_selftest, "MFMA 32x32x16 layout check");
   m.def("conv_tables", &dsin::conv_tables, "gather-conv offset tables");
   m.def("conv_fwd", &dsin::conv_fwd, "implicit-GEMM gather conv forward");
   m.def("conv_wrw", &dsin::conv_wrw, "implicit-GEMM conv weight gradient");

@@ -142,3 +142,41 @@ torch::Tensor mfma_selftest(torch::Tensor A, torch::Tensor B) {
 }
 
 }  // namespace dsin
+
+namespace dsin {
+// 32x32x16 MFMA layout self-check: C = A(32x16) @ B(16x32), one wave.
+__global__ void mfma32_selftest_kernel(const float* __restrict__ A,
+                                       const float* __restrict__ B,
+                                       float* __restrict__ C) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 b8;
+  typedef __attribute__((ext_vector_type(16))) float f16v;
+  int lane = threadIdx.x & 63;
+  b8 a, b;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    int k = (lane >> 5) * 8 + e;
+    ncbf16 av = nf2b(A[(lane & 31) * 16 + k]);   // A[row][k]
+    ncbf16 bv = nf2b(B[k * 32 + (lane & 31)]);   // B[k][col]
+    a[e] = *reinterpret_cast<__bf16*>(&av);
+    b[e] = *reinterpret_cast<__bf16*>(&bv);
+  }
+  f16v acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+    int col = lane & 31;
+    C[row * 32 + col] = acc[reg];
+  }
+}
+
+torch::Tensor mfma32_selftest(torch::Tensor A, torch::Tensor B) {
+  CHECK_CUDA_CONTIG(A);
+  CHECK_CUDA_CONTIG(B);
+  auto C = torch::empty({32, 32}, A.options());
+  hipLaunchKernelGGL(mfma32_selftest_kernel, dim3(1), dim3(64), 0,
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(), A.data_ptr<float>(),
+                     B.data_ptr<float>(), C.data_ptr<float>());
+  return C;
+}
+}  // namespace dsin
