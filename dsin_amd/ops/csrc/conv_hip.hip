@@ -44,7 +44,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
     bptr = bias->data_ptr<float>();
   }
   dim3 grid((M + CONV_TM - 1) / CONV_TM, (N + CONV_TN - 1) / CONV_TN, B);
-  size_t lds = (size_t)2 * CONV_TM * (64 + CONV_AP) * 2;
+  size_t lds = (size_t)4 * CONV_TM * (64 + CONV_AP) * 2;  // 4-buffer pipeline
   hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(256), lds,
                      at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                      (const cvbf16*)xbuf.data_ptr(),

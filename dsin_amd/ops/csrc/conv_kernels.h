@@ -175,37 +175,77 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
       }
   };
 
-  // Depth-1 pipeline: 2 LDS buffers; the next chunk's W fragments and A
-  // values are loaded to registers while the current chunk's MFMAs run;
-  // the register->LDS write lands after the MFMA phase, one barrier per
-  // chunk. (A 2-deep variant showed rare memory faults under stress —
-  // kept out until root-caused; see tests/test_gpu_conv.py stress test.)
+  // 2-deep software pipeline: 4 LDS buffers, two register stage sets.
+  // Invariant at iteration kt (even): buf[kt&3] holds chunk kt, set S1
+  // holds chunk kt+1; the body issues loads for chunk kt+2 into S0, MFMAs
+  // chunk kt, writes S1 into buf[(kt+1)&3], barriers; odd iterations swap
+  // the sets. A-prefetch therefore has ~two MFMA phases to land.
   const int nchunks = KP / KC;
-
-  load_chunk(0);
-  write_chunk(0);
-  cv_bf16x8 wcur[2] = {wfrag[0], wfrag[1]};
-  __syncthreads();
-
-  for (int kt = 0; kt < nchunks; ++kt) {
-    if (kt + 1 < nchunks) load_chunk((kt + 1) * KC);
-    const char* cur = As8 + (kt & 1) * ABUF * 2;
+  u16x8 stageB[2];
+  cv_bf16x8 wfragB[2];
+  auto load_chunkB = [&](int kc) {
+    wfragB[0] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + kgrp * 8]);
+    wfragB[1] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + 32 + kgrp * 8]);
+    load_half(kc + sk, stageB[0]);
+    load_half(kc + 32 + sk, stageB[1]);
+  };
+  auto write_chunkB = [&](int buf) {
+    char* dst = As8 + buf * ABUF * 2;
+#pragma unroll
+    for (int h = 0; h < 2; ++h)
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        unsigned short u = stageB[h][i];
+        *reinterpret_cast<cvbf16*>(&dst[aswz(sm8 + i, h * 32 + sk)]) =
+            *reinterpret_cast<cvbf16*>(&u);
+      }
+  };
+  auto mfma_chunk = [&](int kt, const cv_bf16x8* w2) {
+    const char* cur = As8 + (kt & 3) * ABUF * 2;
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
             &cur[aswz(mi * 16 + colL, kk * 32 + kgrp * 8)]);
-        acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, wcur[kk],
+        acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, w2[kk],
                                                           acc[mi], 0, 0, 0);
       }
     }
-    if (kt + 1 < nchunks) {
-      wcur[0] = wfrag[0];   // W(kt+1): copied after the MFMA phase so its
-      wcur[1] = wfrag[1];   // wait never stalls the matrix pipe
-      write_chunk((kt + 1) & 1);
+  };
+
+  load_chunk(0);              // -> stage/wfrag (set A)
+  write_chunk(0);
+  cv_bf16x8 wA[2] = {wfrag[0], wfrag[1]};
+  cv_bf16x8 wB[2] = {wA[0], wA[1]};
+  if (nchunks > 1) {
+    load_chunkB(KC);          // chunk 1 -> set B
+    wB[0] = wfragB[0];
+    wB[1] = wfragB[1];
+  }
+  __syncthreads();
+
+  for (int kt = 0; kt + 1 < nchunks; kt += 2) {
+    // even iteration: compute chunk kt (in buf kt&3), set B holds kt+1
+    if (kt + 2 < nchunks) load_chunk((kt + 2) * KC);
+    mfma_chunk(kt, wA);
+    write_chunkB((kt + 1) & 3);
+    __syncthreads();
+    // odd iteration: compute chunk kt+1, set A holds kt+2
+    if (kt + 3 < nchunks) load_chunkB((kt + 3) * KC);
+    mfma_chunk(kt + 1, wB);
+    wA[0] = wfrag[0];   // W(kt+2): copied only after both MFMA phases so
+    wA[1] = wfrag[1];   // its wait never stalls the matrix pipe
+    if (kt + 2 < nchunks) {
+      write_chunk((kt + 2) & 3);
       __syncthreads();
     }
+    wB[0] = wfragB[0];
+    wB[1] = wfragB[1];
+  }
+  if (nchunks & 1) {
+    // odd count: last chunk sits in set A's buffer (written above)
+    mfma_chunk(nchunks - 1, wA);
   }
 
   // ---- epilogue: D[row=pixel][col=cout]; row = mi*16 + kgrp*4 + reg ----
