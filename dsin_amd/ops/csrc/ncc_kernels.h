@@ -36,11 +36,10 @@ __constant__ float SIF_STD[3] = {73.56493292844912f, 75.88547006820752f,
 constexpr float FOURLN2 = 2.772588722239781f;  // 4 ln 2
 constexpr float NCC_EPS = 1e-10f;  // guards zero-variance windows (both paths)
 
-constexpr int NCC_TP = 16;   // patches per workgroup (MFMA M tile)
-constexpr int NCC_TJ = 64;   // cols per workgroup (4 waves x 16)
+constexpr int NCC_TP = 32;   // patches per workgroup (MFMA 32x32 M tile)
+constexpr int NCC_TJ = 128;  // cols per workgroup (4 waves x 32)
 constexpr int NCC_TI = 8;    // rows looped per workgroup
-constexpr int NCC_APAD = 8;  // bf16 pad per A row -> 16 distinct banks on the
-                             // column-wise ds_read_b128 (stride 20 dwords)
+constexpr int NCC_APAD = 8;  // bf16 pad per A row for the b128 column reads
 
 // ---------------------------------------------------------------- stage 0
 // Offset tables, computed once per call:
@@ -159,8 +158,12 @@ void ncc_main_kernel(const ncbf16* __restrict__ tx, const ncbf16* __restrict__ t
                      unsigned long long* __restrict__ best,  // (P,)
                      int H, int W, int ph, int pw, int gw, int P,
                      int Hc, int Wc, int use_mask) {
+  // mfma_f32_32x32x16_bf16: A[row=l&31][k=(l>>5)*8+e], B[k][col=l&31],
+  // C row = (reg&3) + 8*(reg>>2) + 4*(l>>5). 2x the flops per B-gather of
+  // the 16x16 shape (the gather is the cost here). Two interleaved i-rows
+  // keep two independent accumulator chains in flight.
   const int K = 3 * ph * pw;
-  const int KP = (K + 31) & ~31;
+  const int KP = (K + 15) & ~15;
   const int YR = NCC_TI + ph - 1;
   const int YC = NCC_TJ + pw - 1;
   const int YCP = (YC + 8) & ~7;
@@ -170,6 +173,7 @@ void ncc_main_kernel(const ncbf16* __restrict__ tx, const ncbf16* __restrict__ t
   ncbf16* As = reinterpret_cast<ncbf16*>(smem);                // [TP][ASTRIDE]
   ncbf16* Ys = As + NCC_TP * ASTRIDE;                          // [3][YR][YCP]
   unsigned short* Ko = reinterpret_cast<unsigned short*>(Ys + 3 * YR * YCP);
+  float* Pstat = reinterpret_cast<float*>(Ko + ((KP + 7) & ~7));  // [5][TP]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -178,9 +182,23 @@ void ncc_main_kernel(const ncbf16* __restrict__ tx, const ncbf16* __restrict__ t
   const int i0 = blockIdx.y * NCC_TI;
   const int p0 = blockIdx.z * NCC_TP;
 
-  // ---- stage LDS: A tile (16 patches x KP), y window, koffs copy ------
-  for (int k = tid; k < KP; k += 256) {
-    Ko[k] = (k < K) ? koffs[k] : (unsigned short)0;
+  const float fK = (float)K;
+  const float invK = 1.f / fK;
+  const float ish2 = 4.0f / ((float)H * (float)H);   // 1/sh^2, sh = H/2
+  const float isw2 = 4.0f / ((float)W * (float)W);
+
+  // ---- stage A tile (32 patches x K), y window, koffs, patch stats ----
+  const int php_w = ph * pw;
+  for (int k = tid; k < KP; k += 256) Ko[k] = (k < K) ? koffs[k] : 0;
+  if (tid < NCC_TP) {
+    const int p = min(p0 + tid, P - 1);
+    const float sxv = psum[p];
+    const float xm = sxv * invK;
+    Pstat[tid] = sxv;
+    Pstat[NCC_TP + tid] = xm;
+    Pstat[2 * NCC_TP + tid] = psum2[p] - 2.f * xm * sxv + fK * xm * xm;
+    Pstat[3 * NCC_TP + tid] = ((float)(p / gw) + 0.5f) * (float)ph;  // cr
+    Pstat[4 * NCC_TP + tid] = ((float)(p % gw) + 0.5f) * (float)pw;  // cw
   }
 #pragma unroll 1
   for (int pi = 0; pi < NCC_TP; ++pi) {
@@ -208,94 +226,92 @@ void ncc_main_kernel(const ncbf16* __restrict__ tx, const ncbf16* __restrict__ t
   }
   __syncthreads();
 
-  const int colL = lane & 15;
-  const int kgrp = lane >> 4;
-  const int jj = wid * 16 + colL;
+  const int colL = lane & 31;          // col within the wave's 32
+  const int kgrp = lane >> 5;          // 0/1 (k-offset 0/8 within 16)
+  const int jj = wid * 32 + colL;
   const int j = j0 + jj;
+  const bool jvalid = j < Wc;
 
-  const float fK = (float)K;
-  const float invK = 1.f / fK;
-  const float ish2 = 4.0f / ((float)H * (float)H);   // 1/sh^2, sh = H/2
-  const float isw2 = 4.0f / ((float)W * (float)W);
-
-  // per-reg patch constants (i-independent): stats + prior centers
-  float sxv[4], xm[4], denx[4], cr[4], dj2w[4];
-  bool pvalid[4];
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  unsigned long long bestk[16];
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    const int prow = p0 + kgrp * 4 + reg;
-    pvalid[reg] = prow < P;
-    const int pc = pvalid[reg] ? prow : 0;
-    sxv[reg] = psum[pc];
-    xm[reg] = sxv[reg] * invK;
-    denx[reg] = psum2[pc] - 2.f * xm[reg] * sxv[reg] + fK * xm[reg] * xm[reg];
-    cr[reg] = ((float)(pc / gw) + 0.5f) * (float)ph;
-    const float cw = ((float)(pc % gw) + 0.5f) * (float)pw;
-    const float dj = (float)(j + pw / 2 - 1) - cw;
-    dj2w[reg] = dj * dj * isw2;
-  }
+  for (int r = 0; r < 16; ++r) bestk[r] = 0ull;
 
-  unsigned long long bestk[4] = {0ull, 0ull, 0ull, 0ull};
-  const ncbf16* __restrict__ arow = &As[colL * ASTRIDE + kgrp * 8];
+  const ncbf16* __restrict__ arow = &As[kgrp * 8];
   const unsigned short* __restrict__ krow = &Ko[kgrp * 8];
+  typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
 
-  for (int i = 0; i < NCC_TI; ++i) {
+  for (int i = 0; i < NCC_TI; i += 2) {
     const int ii = i0 + i;
     if (ii >= Hc) break;
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    const ncbf16* __restrict__ ybase = &Ys[i * YCP + jj];
-    for (int kt = 0; kt < KP / 32; ++kt) {
-      const bf16x8 afrag =
-          *reinterpret_cast<const bf16x8*>(arow + kt * 32);
-      // 8 broadcast offsets (one b128 across the 16-lane group)...
-      typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+    const bool i1ok = (ii + 1) < Hc;
+    f32x16 acc0 = {};
+    f32x16 acc1 = {};
+    const ncbf16* __restrict__ yb0 = &Ys[i * YCP + jj];
+    const ncbf16* __restrict__ yb1 = &Ys[(i + 1) * YCP + jj];
+    for (int kt = 0; kt < KP / 16; ++kt) {
       const u16x8 ko8 =
-          *reinterpret_cast<const u16x8*>(krow + kt * 32);
-      bf16x8 bfrag;
+          *reinterpret_cast<const u16x8*>(krow + kt * 16);
+      bf16x8 bf0, bf1;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        ncbf16 v = ybase[ko8[e]];          // ...then 8 u16 LDS value reads
-        bfrag[e] = *reinterpret_cast<__bf16*>(&v);
+        ncbf16 v0 = yb0[ko8[e]];
+        ncbf16 v1 = yb1[ko8[e]];
+        bf0[e] = *reinterpret_cast<__bf16*>(&v0);
+        bf1[e] = *reinterpret_cast<__bf16*>(&v1);
       }
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc, 0, 0, 0);
+      // A fragment: row = colL (patch), k = kgrp*8 + e
+      const bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+          &arow[colL * ASTRIDE + kt * 16]);
+      acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, bf0, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(afrag, bf1, acc1, 0, 0, 0);
     }
-    // ---- predicated epilogue ----
-    const bool jvalid = j < Wc;
-    const long long sidx = (long long)ii * Wc + (jvalid ? j : 0);
-    const float syv = sy[sidx];
-    const float sy2v = sy2[sidx];
-    const float ym = syv * invK;
-    const float deny = sy2v - 2.f * ym * syv + fK * ym * ym;
-    const float di = (float)(ii + ph / 2 - 1);
+    // ---- epilogue for rows ii and ii+1 ----
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      const float num = acc[reg] - ym * sxv[reg] - xm[reg] * syv
-                        + fK * xm[reg] * ym;
-      float val = num * __builtin_amdgcn_rsqf(
-                            fmaxf(denx[reg] * deny, NCC_EPS));
-      if (use_mask) {
-        const float dd = di - cr[reg];
-        val *= __expf(-FOURLN2 * (dd * dd * ish2 + dj2w[reg]));
+    for (int half = 0; half < 2; ++half) {
+      if (half == 1 && !i1ok) break;
+      const int iw = ii + half;
+      const long long sidx = (long long)iw * Wc + (jvalid ? j : 0);
+      const float syv = sy[sidx];
+      const float sy2v = sy2[sidx];
+      const float ym = syv * invK;
+      const float deny = sy2v - 2.f * ym * syv + fK * ym * ym;
+      const float di = (float)(iw + ph / 2 - 1);
+      const unsigned int idx = (unsigned int)(iw * Wc + j);
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int prow = (reg & 3) + 8 * (reg >> 2) + 4 * kgrp;
+        const float av = half == 0 ? acc0[reg] : acc1[reg];
+        const float sxv = Pstat[prow];
+        const float xm = Pstat[NCC_TP + prow];
+        const float denx = Pstat[2 * NCC_TP + prow];
+        const float num = av - ym * sxv - xm * syv + fK * xm * ym;
+        float val = num * __builtin_amdgcn_rsqf(fmaxf(denx * deny, NCC_EPS));
+        if (use_mask) {
+          const float dd = di - Pstat[3 * NCC_TP + prow];
+          const float dj = (float)(j + pw / 2 - 1) - Pstat[4 * NCC_TP + prow];
+          val *= __expf(-FOURLN2 * (dd * dd * ish2 + dj * dj * isw2));
+        }
+        unsigned long long key =
+            ((unsigned long long)nfloat_flip(val) << 32) |
+            (unsigned long long)(~idx);
+        key = (jvalid && (p0 + prow) < P) ? key : 0ull;
+        if (key > bestk[reg]) bestk[reg] = key;
       }
-      const unsigned int idx = (unsigned int)(ii * Wc + j);
-      unsigned long long key = ((unsigned long long)nfloat_flip(val) << 32) |
-                               (unsigned long long)(~idx);
-      key = (jvalid && pvalid[reg]) ? key : 0ull;
-      if (key > bestk[reg]) bestk[reg] = key;
     }
   }
 
-  // reduce across the 16 lanes of each kgrp group (distinct cols, same rows)
+  // reduce across the 32 lanes of each half (distinct cols, same row set)
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
+  for (int reg = 0; reg < 16; ++reg) {
     unsigned long long k = bestk[reg];
 #pragma unroll
-    for (int off = 8; off > 0; off >>= 1) {
+    for (int off = 16; off > 0; off >>= 1) {
       unsigned long long other =
-          (unsigned long long)__shfl_xor((long long)k, off, 16);
+          (unsigned long long)__shfl_xor((long long)k, off, 32);
       if (other > k) k = other;
     }
-    const int prow = p0 + kgrp * 4 + reg;
+    const int prow = p0 + (reg & 3) + 8 * (reg >> 2) + 4 * kgrp;
     if (colL == 0 && prow < P && k != 0ull) atomicMax(&best[prow], k);
   }
 }

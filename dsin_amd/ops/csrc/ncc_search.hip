@@ -37,10 +37,11 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ncc_search(
   const int gh = H / ph, gw = W / pw, P = gh * gw;
   const int Hc = H - ph + 1, Wc = W - pw + 1;
   const int K = 3 * ph * pw;
-  const int KP = (K + 31) & ~31;
+  const int KP = (K + 15) & ~15;
   const int YR = NCC_TI + ph - 1, YC = NCC_TJ + pw - 1, YCP = (YC + 8) & ~7;
   const size_t lds = (size_t)NCC_TP * (KP + NCC_APAD) * 2 +
-                     (size_t)3 * YR * YCP * 2 + (size_t)KP * 2;
+                     (size_t)3 * YR * YCP * 2 + (size_t)((KP + 7) & ~7) * 2 +
+                     (size_t)5 * NCC_TP * 4;
   TORCH_CHECK(lds <= 160 * 1024, "patch size too large for LDS tiling: ", lds);
   TORCH_CHECK(3 * YR * YCP < 65536, "y-window exceeds u16 offset range");
 
