@@ -45,6 +45,10 @@ def main(argv=None):
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--bf16", action="store_true", help="bf16 autocast compute")
     ap.add_argument("--metrics", type=str, default=None, help="JSONL metrics path")
+    ap.add_argument("--checkpoint-every", type=int, default=0,
+                    help="periodic crash-recovery checkpoint interval "
+                         "(iterations; 0 = best-val only, the reference's "
+                         "behavior)")
     args = ap.parse_args(argv)
 
     ae_config, _ = config_mod.parse(args.ae_config_path)
@@ -111,6 +115,13 @@ def main(argv=None):
                                         root_weights, model_name, iteration,
                                         total_iterations, best_val, ae_config, pc_config)
                         print(f"Saved {root_weights + model_name}")
+
+            if (args.checkpoint_every and rank() == 0
+                    and iteration % args.checkpoint_every == 0):
+                checkpoint.save(model, trainer.optimizers, trainer.global_step,
+                                root_weights, "periodic", iteration,
+                                total_iterations, float(val_loss),
+                                ae_config, pc_config)
 
             if iteration % show_every == 0:
                 print(f"[{iteration}/{total_iterations}] loss {train_sum/show_every:.4f} "
