@@ -154,7 +154,9 @@ class Trainer:
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
                 for _ in range(2):
+                    # real parameter updates on the capture input
                     self._step_inner(self._static_x, self._static_y)
+                    self.global_step += 1
             torch.cuda.current_stream().wait_stream(side)
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
