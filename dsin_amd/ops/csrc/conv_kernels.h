@@ -163,6 +163,20 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
     return ((m * ASTR + elem_off) * 2) ^ (((m >> 3) & 7) << 4);
   };
   char* As8 = reinterpret_cast<char*>(As);
+  // chunk-invariant offsets, computed once (the staging/read address math
+  // was ~12 VALU per MFMA when recomputed per chunk)
+  int wr_off[16];
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      wr_off[h * 8 + i] = aswz(sm8 + i, h * 32 + sk);
+  int rd_off[8];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      rd_off[kk * 4 + mi] = aswz(mi * 16 + colL, kk * 32 + kgrp * 8);
   auto write_chunk = [&](int buf) {
     char* dst = As8 + buf * ABUF * 2;
 #pragma unroll
@@ -170,7 +184,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         unsigned short u = stage[h][i];
-        *reinterpret_cast<cvbf16*>(&dst[aswz(sm8 + i, h * 32 + sk)]) =
+        *reinterpret_cast<cvbf16*>(&dst[wr_off[h * 8 + i]]) =
             *reinterpret_cast<cvbf16*>(&u);
       }
   };
@@ -196,7 +210,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         unsigned short u = stageB[h][i];
-        *reinterpret_cast<cvbf16*>(&dst[aswz(sm8 + i, h * 32 + sk)]) =
+        *reinterpret_cast<cvbf16*>(&dst[wr_off[h * 8 + i]]) =
             *reinterpret_cast<cvbf16*>(&u);
       }
   };
@@ -207,7 +221,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
-            &cur[aswz(mi * 16 + colL, kk * 32 + kgrp * 8)]);
+            &cur[rd_off[kk * 4 + mi]]);
         acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, w2[kk],
                                                           acc[mi], 0, 0, 0);
       }
