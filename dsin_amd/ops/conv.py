@@ -94,15 +94,6 @@ def _padded_buf(x: torch.Tensor, pl: int, pr: int, pt: int, pb: int,
     return buf
 
 
-def _zero_stuff(x: torch.Tensor, stride: int) -> torch.Tensor:
-    if stride == 1:
-        return x
-    b, c, h, w = x.shape
-    z = x.new_zeros(b, c, (h - 1) * stride + 1, (w - 1) * stride + 1)
-    z[..., ::stride, ::stride] = x
-    return z
-
-
 class _GatherConvFn(torch.autograd.Function):
     """y[b, co, oh, ow] = sum_{ci,r,s} xbuf[b, ci, oh*st + r*d, ow*st + s*d]
     * w1[co, (ci, r, s)]  (+bias, +activation). VALID over xbuf."""

@@ -29,9 +29,6 @@ __all__ = [
     "gaussian_mask_value",
 ]
 
-_HARD = False  # placeholder
-
-
 # ---------------------------------------------------------------------------
 # Quantizer (reference src/quantizer_imgcomp.py:37-100 and the straight-through
 # combine at src/autoencoder_imgcomp.py:127-134)
