@@ -52,7 +52,8 @@ def main():
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--ae-only", action="store_true",
                     help="benchmark config 2 (AE without side information)")
-    ap.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--dtype", type=str, default="bf16",
+                    choices=["bf16", "fp32", "fp8"])
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph capture of the train step")
@@ -70,6 +71,9 @@ def main():
     n_ranks = world_size()
 
     ae_config, pc_config = build_configs(args)
+    if args.dtype == "fp8":
+        from dsin_amd.ops import conv as dconv
+        dconv.set_compute_dtype("fp8")  # e4m3 MFMA conv path (config 5)
     torch.manual_seed(1234 + rank())
     model = DSIN(ae_config, pc_config).to(device)
     if args.channels_last:
@@ -82,7 +86,7 @@ def main():
     use_graph = (not args.no_graph and device.type == "cuda"
                  and world_size() == 1)
     trainer = Trainer(model, ae_config, pc_config, num_training_imgs=1576,
-                     device=device, autocast_bf16=(args.dtype == "bf16"),
+                     device=device, autocast_bf16=(args.dtype in ("bf16", "fp8")),
                      use_cuda_graph=use_graph,
                      ddp_comm_dtype=None)
 

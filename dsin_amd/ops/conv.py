@@ -30,6 +30,20 @@ from . import _require_ext, hip_available
 
 _PLANS = {}
 
+# compute dtype for the conv path: "bf16" (default) or "fp8" (OCP e4m3,
+# BASELINE config 5). fp8 tensors are carried as raw-byte uint8 views.
+_COMPUTE = "bf16"
+
+
+def set_compute_dtype(dtype: str) -> None:
+    global _COMPUTE
+    assert dtype in ("bf16", "fp8")
+    _COMPUTE = dtype
+
+
+def compute_dtype() -> str:
+    return _COMPUTE
+
 
 def _plan(device, Ci: int, Hp: int, Wp: int, kh: int, kw: int, stride: int,
           dil: int, HO: int, WO: int):
@@ -46,10 +60,16 @@ def _plan(device, Ci: int, Hp: int, Wp: int, kh: int, kw: int, stride: int,
 
 
 def _wmat(w1: torch.Tensor) -> torch.Tensor:
-    """(Co, K) any-dtype -> bf16 zero-padded to (Co, KP64+8); the zeros
-    cancel the clamped out-of-range A gathers in the kernel."""
+    """(Co, K) any-dtype -> bf16 (or e4m3-as-uint8) zero-padded to
+    (Co, KP64+8); the zeros cancel the clamped out-of-range A gathers."""
     K = w1.shape[1]
     KP = (K + 63) & ~63
+    if _COMPUTE == "fp8" and w1.is_cuda:
+        w8 = w1.to(torch.float8_e4m3fn).view(torch.uint8)
+        out = torch.zeros(w1.shape[0], KP + 8, dtype=torch.uint8,
+                          device=w1.device)
+        out[:, :K] = w8
+        return out
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
 
 
@@ -61,7 +81,7 @@ class _PadStuffFn(torch.autograd.Function):
     def forward(ctx, x, pt, pb, pl, pr, stride):
         fn = _require_ext("pad_stuff")
         ctx.meta = (pt, pl, stride, x.shape, x.dtype)
-        return fn(x.contiguous(), pt, pb, pl, pr, stride)
+        return fn(x.contiguous(), pt, pb, pl, pr, stride, _COMPUTE == "fp8")
 
     @staticmethod
     def backward(ctx, dy):
@@ -69,6 +89,8 @@ class _PadStuffFn(torch.autograd.Function):
         B, C, H, W = shape
         dx = dy[:, :, pt:pt + (H - 1) * stride + 1:stride,
                 pl:pl + (W - 1) * stride + 1:stride]
+        if dx.dtype == torch.uint8:  # e4m3 bytes from the fp8 path
+            dx = dx.view(torch.float8_e4m3fn).to(torch.float32)
         return dx.to(dtype), None, None, None, None, None
 
 
@@ -126,6 +148,9 @@ class _GatherConvFn(torch.autograd.Function):
                                                         device=dy.device))
         elif act == 2:
             dy = torch.where(y_act > 0, dy, dy * 0.2)
+        fp8 = xbuf.dtype == torch.uint8
+        dy_k = (dy.to(torch.float8_e4m3fn).view(torch.uint8).contiguous()
+                if fp8 else dy)
 
         dxbuf = None
         if ctx.needs_input_grad[0]:
@@ -143,7 +168,7 @@ class _GatherConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil,
                                 HO, WO)
-            dw1 = ext_wrw(xbuf, dy, mbase, koff, Co, K, WO,
+            dw1 = ext_wrw(xbuf, dy_k, mbase, koff, Co, K, WO,
                           stride == 1).to(w1.dtype)
 
         dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None

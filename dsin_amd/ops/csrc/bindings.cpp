@@ -31,7 +31,7 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor lr, torch::Tensor step,
                double b1, double b2, double eps);
 torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
-                        int64_t pr, int64_t stride);
+                        int64_t pr, int64_t stride, bool fp8);
 std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
                                   torch::Tensor beta, torch::Tensor rmean,
                                   torch::Tensor rvar, double momentum,

@@ -5,6 +5,7 @@
 
 #include "common_hip.h"
 #include "conv_kernels.h"
+#include "conv_fp8.h"
 
 namespace dsin {
 
@@ -30,20 +31,34 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        int64_t stride) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(wmat);
-  TORCH_CHECK(xbuf.scalar_type() == torch::kBFloat16, "xbuf must be bf16");
-  TORCH_CHECK(wmat.scalar_type() == torch::kBFloat16, "wmat must be bf16");
+  const bool fp8 = xbuf.scalar_type() == torch::kByte;  // raw e4m3 bytes
+  TORCH_CHECK(fp8 || xbuf.scalar_type() == torch::kBFloat16,
+              "xbuf must be bf16 or e4m3-as-uint8");
+  TORCH_CHECK(wmat.scalar_type() == xbuf.scalar_type(), "x/w dtype mismatch");
   const int64_t B = xbuf.size(0);
   const int64_t M = HO * WO;
   const int64_t KP = (K + 63) & ~63;  // 64-chunk padded; wmat zero-padded
   TORCH_CHECK(wmat.size(1) == KP + CONV_AP, "wmat row stride mismatch");
   if (WO < 8) stride = 0;  // scalar staging path for very narrow outputs
-  auto out = torch::empty({B, N, HO, WO}, xbuf.options());
+  auto out = torch::empty({B, N, HO, WO},
+                          xbuf.options().dtype(torch::kBFloat16));
   const float* bptr = nullptr;
   if (bias.has_value()) {
     CHECK_CUDA_CONTIG(bias.value());
     bptr = bias->data_ptr<float>();
   }
   dim3 grid((M + CONV_TM - 1) / CONV_TM, (N + CONV_TN - 1) / CONV_TN, B);
+  if (fp8) {
+    size_t lds = (size_t)2 * CONV8_TM * (64 + CONV8_AP);
+    hipLaunchKernelGGL(conv_fwd_fp8_kernel, grid, dim3(256), lds,
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                       (const f8*)xbuf.data_ptr(), (const f8*)wmat.data_ptr(),
+                       bptr, (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
+                       koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
+                       xbuf.stride(0), (long long)N * M, (int)act, (int)WO,
+                       (int)stride);
+    return out;
+  }
   size_t lds = (size_t)4 * CONV_TM * (64 + CONV_AP) * 2;  // 4-buffer pipeline
   hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(256), lds,
                      at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
@@ -61,11 +76,24 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        int64_t K, int64_t WO, bool mcontig) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(dy);
+  const bool fp8 = xbuf.scalar_type() == torch::kByte;
   const int64_t B = xbuf.size(0);
   const int64_t M = dy.size(2) * dy.size(3);
   auto dw = torch::zeros({N, K}, xbuf.options().dtype(torch::kFloat32));
   int pix_chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 2048, 1), 64);
   dim3 grid((K + 63) / 64, (N + 63) / 64, B * pix_chunks);
+  if (fp8) {
+    TORCH_CHECK(dy.scalar_type() == torch::kByte, "fp8 wrw: dy must be e4m3");
+    size_t lds = (size_t)4 * 64 * (32 + CONV8_AP);
+    hipLaunchKernelGGL(conv_wrw_fp8_kernel, grid, dim3(256), lds,
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                       (const f8*)xbuf.data_ptr(), (const f8*)dy.data_ptr(),
+                       dw.data_ptr<float>(), mbase.data_ptr<int>(),
+                       koff.data_ptr<int>(), (int)M, (int)N, (int)K,
+                       xbuf.stride(0), dy.stride(0), pix_chunks, (int)WO,
+                       (int)(mcontig ? 1 : 0));
+    return dw;
+  }
   size_t lds = (size_t)4 * 64 * (32 + CONV_AP) * 2;  // 2 tiles x dbuf
   hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), lds,
                      at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),

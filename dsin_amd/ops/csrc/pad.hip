@@ -6,6 +6,7 @@
 // vector reads (see ops/conv.py).
 
 #include "common.h"
+#include "conv_fp8.h"
 
 namespace dsin {
 
@@ -45,7 +46,7 @@ __global__ void pad_stuff_kernel(const T* __restrict__ x,
 }
 
 torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
-                        int64_t pr, int64_t stride) {
+                        int64_t pr, int64_t stride, bool fp8) {
   CHECK_CUDA_CONTIG(x);
   TORCH_CHECK(x.dim() == 4, "pad_stuff expects NCHW");
   const int B = (int)x.size(0), C = (int)x.size(1);
@@ -54,11 +55,28 @@ torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
   const int Hp = Hs + (int)(pt + pb), Wp = Ws + (int)(pl + pr);
   const long long n_img = (long long)C * Hp * Wp;
   auto store = torch::empty({(int64_t)B * n_img + 16},
-                            x.options().dtype(torch::kBFloat16));
+                            x.options().dtype(fp8 ? torch::kByte
+                                                  : torch::kBFloat16));
   auto out = store.narrow(0, 0, B * n_img).view({B, C, Hp, Wp});
   long long total = (long long)B * n_img;
   int grid = (int)std::min<long long>((total + 255) / 256, 8192);
   auto stream = at::cuda::getCurrentCUDAStream();
+  if (fp8) {
+    if (x.scalar_type() == torch::kFloat32) {
+      hipLaunchKernelGGL((pad_stuff_fp8_kernel<float>), dim3(grid), dim3(256),
+                         0, stream, x.data_ptr<float>(), (f8*)out.data_ptr(),
+                         C, H, W, Hp, Wp, (int)pt, (int)pl, (int)stride, n_img,
+                         (long long)C * H * W, B);
+    } else if (x.scalar_type() == torch::kBFloat16) {
+      hipLaunchKernelGGL((pad_stuff_fp8_kernel<bf16>), dim3(grid), dim3(256),
+                         0, stream, (const bf16*)x.data_ptr(),
+                         (f8*)out.data_ptr(), C, H, W, Hp, Wp, (int)pt,
+                         (int)pl, (int)stride, n_img, (long long)C * H * W, B);
+    } else {
+      TORCH_CHECK(false, "pad_stuff fp8: fp32 or bf16 input only");
+    }
+    return out;
+  }
   if (x.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((pad_stuff_kernel<float>), dim3(grid), dim3(256), 0,
                        stream, x.data_ptr<float>(), (bf16*)out.data_ptr(), C,

@@ -214,3 +214,33 @@ def test_probclass_gpu_matches_cpu(dev):
     pc_g = pc.to(dev)
     bc_gpu = pc_g.bitcost(q.to(dev), sym.to(dev), torch.tensor(0.5, device=dev))
     torch.testing.assert_close(bc_gpu.cpu(), bc_cpu, rtol=0.05, atol=0.05)
+
+
+def test_conv_fp8_path(dev):
+    """fp8 (e4m3) MFMA conv path: forward within fp8 tolerance of fp32
+    torch conv; gradients flow and are finite."""
+    from dsin_amd.ops import conv as dconv
+    dconv.set_compute_dtype("fp8")
+    try:
+        torch.manual_seed(0)
+        x = torch.randn(1, 32, 24, 40, device=dev)
+        w = torch.randn(48, 32, 3, 3, device=dev) / (3 * 32 ** 0.5)
+        x1 = x.clone().requires_grad_(True)
+        w1 = w.clone().requires_grad_(True)
+        y = dconv.conv2d(x1, w1, None, 1, 1, 1)
+        yr = F.conv2d(x, w, None, stride=1, padding=1)
+        # e4m3 has a 3-bit mantissa: ~6% relative element error
+        rel = (y.float() - yr).abs().max() / yr.abs().max()
+        assert float(rel) < 0.15, float(rel)
+        y.sum().backward()
+        assert torch.isfinite(x1.grad.float()).all()
+        assert torch.isfinite(w1.grad.float()).all()
+        # transposed conv on the fp8 path too
+        wt = torch.randn(32, 16, 3, 3, device=dev) * 0.05
+        yt = dconv.conv_transpose2d(x.clone(), wt, None, 2, 1, 1)
+        ytr = F.conv_transpose2d(x, wt, None, stride=2, padding=1,
+                                 output_padding=1)
+        relt = (yt.float() - ytr).abs().max() / ytr.abs().max()
+        assert float(relt) < 0.15, float(relt)
+    finally:
+        dconv.set_compute_dtype("bf16")
