@@ -4,6 +4,7 @@
 
 #include "common.h"
 #include "conv_kernels.h"
+#define DSIN_CONV_FP8_KERNELS
 #include "conv_fp8.h"
 
 namespace dsin {

@@ -29,7 +29,8 @@ constexpr int CONV8_TM = 64;
 constexpr int CONV8_TN = 64;
 constexpr int CONV8_AP = 8;   // fp8 pad per A row (bytes)
 
-static __global__ __launch_bounds__(256)
+#ifdef DSIN_CONV_FP8_KERNELS
+__global__ __launch_bounds__(256)
 void conv_fwd_fp8_kernel(const f8* __restrict__ xpad,   // (Ci, Hp, Wp) e4m3
                          const f8* __restrict__ wmat,   // (Co, KP64+AP) 0-pad
                          const float* __restrict__ bias,
@@ -184,7 +185,7 @@ void conv_fwd_fp8_kernel(const f8* __restrict__ xpad,   // (Ci, Hp, Wp) e4m3
 }
 
 // dW accumulation in fp8 inputs (x and dy both e4m3), fp32 atomics out.
-static __global__ __launch_bounds__(256)
+__global__ __launch_bounds__(256)
 void conv_wrw_fp8_kernel(const f8* __restrict__ xpad, const f8* __restrict__ dy,
                          float* __restrict__ dw,
                          const int* __restrict__ mbase,
@@ -286,6 +287,8 @@ void conv_wrw_fp8_kernel(const f8* __restrict__ xpad, const f8* __restrict__ dy,
     }
   }
 }
+
+#endif  // DSIN_CONV_FP8_KERNELS
 
 // fused pad/stuff/cast-to-e4m3 (fp8 twin of pad_stuff_kernel)
 template <typename T>

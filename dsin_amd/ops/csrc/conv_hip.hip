@@ -5,6 +5,7 @@
 
 #include "common_hip.h"
 #include "conv_kernels.h"
+#define DSIN_CONV_FP8_KERNELS
 #include "conv_fp8.h"
 
 namespace dsin {
