@@ -59,12 +59,12 @@ def _plan(device, Ci: int, Hp: int, Wp: int, kh: int, kw: int, stride: int,
     return p
 
 
-def _wmat(w1: torch.Tensor) -> torch.Tensor:
+def _wmat(w1: torch.Tensor, fp8: bool = False) -> torch.Tensor:
     """(Co, K) any-dtype -> bf16 (or e4m3-as-uint8) zero-padded to
     (Co, KP64+8); the zeros cancel the clamped out-of-range A gathers."""
     K = w1.shape[1]
     KP = (K + 63) & ~63
-    if _COMPUTE == "fp8" and w1.is_cuda:
+    if fp8:
         w8 = w1.to(torch.float8_e4m3fn).view(torch.uint8)
         out = torch.zeros(w1.shape[0], KP + 8, dtype=torch.uint8,
                           device=w1.device)
@@ -81,7 +81,7 @@ class _PadStuffFn(torch.autograd.Function):
     def forward(ctx, x, pt, pb, pl, pr, stride):
         fn = _require_ext("pad_stuff")
         ctx.meta = (pt, pl, stride, x.shape, x.dtype)
-        return fn(x.contiguous(), pt, pb, pl, pr, stride, _COMPUTE == "fp8")
+        return fn(x.contiguous(), pt, pb, pl, pr, stride, False)
 
     @staticmethod
     def backward(ctx, dy):
@@ -89,8 +89,6 @@ class _PadStuffFn(torch.autograd.Function):
         B, C, H, W = shape
         dx = dy[:, :, pt:pt + (H - 1) * stride + 1:stride,
                 pl:pl + (W - 1) * stride + 1:stride]
-        if dx.dtype == torch.uint8:  # e4m3 bytes from the fp8 path
-            dx = dx.view(torch.float8_e4m3fn).to(torch.float32)
         return dx.to(dtype), None, None, None, None, None
 
 
@@ -148,10 +146,6 @@ class _GatherConvFn(torch.autograd.Function):
                                                         device=dy.device))
         elif act == 2:
             dy = torch.where(y_act > 0, dy, dy * 0.2)
-        fp8 = xbuf.dtype == torch.uint8
-        dy_k = (dy.to(torch.float8_e4m3fn).view(torch.uint8).contiguous()
-                if fp8 else dy)
-
         dxbuf = None
         if ctx.needs_input_grad[0]:
             with torch.no_grad():
@@ -168,11 +162,82 @@ class _GatherConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil,
                                 HO, WO)
-            dw1 = ext_wrw(xbuf, dy_k, mbase, koff, Co, K, WO,
+            dw1 = ext_wrw(xbuf, dy, mbase, koff, Co, K, WO,
                           stride == 1).to(w1.dtype)
 
         dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
         return dxbuf, dw1, dbias, None, None, None, None, None, None, None
+
+
+class _GatherConvFP8Fn(torch.autograd.Function):
+    """fp8 (e4m3) twin of _GatherConvFn that owns the pad/stuff step itself:
+    fp8 buffers are integer-typed (raw bytes) and cannot carry autograd, so
+    the differentiable boundary is the ORIGINAL float x. pads = (pl, pr, pt,
+    pb); stuff = zero-stuffing stride of the input (conv-transpose)."""
+
+    @staticmethod
+    def forward(ctx, x, w1, bias, stride, dil, kh, kw, HO, WO, act, pads,
+                stuff):
+        pad_fn = _require_ext("pad_stuff")
+        ext_fwd = _require_ext("conv_fwd")
+        pl, pr, pt, pb = pads
+        xbuf = pad_fn(x.contiguous(), pt, pb, pl, pr, stuff, True)
+        B, Ci, Hp, Wp = xbuf.shape
+        Co, K = w1.shape
+        mbase, koff = _plan(x.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
+        bias32 = bias.float().contiguous() if bias is not None else None
+        y = ext_fwd(xbuf, _wmat(w1, fp8=True), bias32, mbase, koff, Co, K,
+                    HO, WO, act, stride)
+        ctx.save_for_backward(xbuf, w1, y if act else None)
+        ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act,
+                    pads, stuff, x.shape, x.dtype)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        xbuf, w1, y_act = ctx.saved_tensors
+        (stride, dil, kh, kw, HO, WO, has_bias, act, pads, stuff, xshape,
+         xdtype) = ctx.meta
+        pad_fn = _require_ext("pad_stuff")
+        ext_fwd = _require_ext("conv_fwd")
+        ext_wrw = _require_ext("conv_wrw")
+        B, Ci, Hp, Wp = xbuf.shape
+        Co, K = w1.shape
+        pl, pr, pt, pb = pads
+
+        dy = dy.contiguous().to(torch.bfloat16)
+        if act == 1:
+            dy = torch.where(y_act > 0, dy, torch.zeros((), dtype=dy.dtype,
+                                                        device=dy.device))
+        elif act == 2:
+            dy = torch.where(y_act > 0, dy, dy * 0.2)
+        dy8 = pad_fn(dy, 0, 0, 0, 0, 1, True).view(B, Co, HO, WO)
+
+        dx = None
+        if ctx.needs_input_grad[0]:
+            with torch.no_grad():
+                pe_h, pe_w = (kh - 1) * dil, (kw - 1) * dil
+                dybuf = pad_fn(dy, pe_h, pe_h, pe_w, pe_w, stride, True)
+                wrot = (w1.view(Co, Ci, kh, kw).flip(2, 3).permute(1, 0, 2, 3)
+                        .reshape(Ci, Co * kh * kw))
+                mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
+                                 kh, kw, 1, dil, Hp, Wp)
+                dxbuf = ext_fwd(dybuf, _wmat(wrot, fp8=True), None, mb2, ko2,
+                                Ci, Co * kh * kw, Hp, Wp, 0, 1)
+                _, _, H, W = xshape
+                dx = dxbuf[:, :, pt:pt + (H - 1) * stuff + 1:stuff,
+                           pl:pl + (W - 1) * stuff + 1:stuff].to(xdtype)
+
+        dw1 = None
+        if ctx.needs_input_grad[1]:
+            mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil,
+                                HO, WO)
+            dw1 = ext_wrw(xbuf, dy8, mbase, koff, Co, K, WO,
+                          stride == 1).to(w1.dtype)
+
+        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        return (dx, dw1, dbias, None, None, None, None, None, None, None,
+                None, None)
 
 
 def _out_size(h: int, k: int, s: int, p: int, d: int) -> int:
@@ -191,6 +256,10 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
     Co, _, kh, kw = w.shape
     HO = _out_size(H, kh, stride, padding, dilation)
     WO = _out_size(W, kw, stride, padding, dilation)
+    if _COMPUTE == "fp8":
+        return _GatherConvFP8Fn.apply(x, w.reshape(Co, Ci * kh * kw), bias,
+                                      stride, dilation, kh, kw, HO, WO, act,
+                                      (padding, padding, padding, padding), 1)
     xbuf = _padded_buf(x, padding, padding, padding, padding)
     return _GatherConvFn.apply(xbuf, w.reshape(Co, Ci * kh * kw), bias,
                                stride, dilation, kh, kw, HO, WO, act)
@@ -212,9 +281,13 @@ def conv_transpose2d(x: torch.Tensor, w: torch.Tensor,
     HO = (H - 1) * stride - 2 * padding + kh + output_padding
     WO = (W - 1) * stride - 2 * padding + kw + output_padding
     pl_h, pl_w = kh - 1 - padding, kw - 1 - padding
+    w1 = w.flip(2, 3).permute(1, 0, 2, 3).reshape(Co, Ci * kh * kw)
+    if _COMPUTE == "fp8":
+        return _GatherConvFP8Fn.apply(
+            x, w1, bias, 1, 1, kh, kw, HO, WO, act,
+            (pl_w, pl_w + output_padding, pl_h, pl_h + output_padding), stride)
     xbuf = _padded_buf(x, pl_w, pl_w + output_padding,
                        pl_h, pl_h + output_padding, stride=stride)
-    w1 = w.flip(2, 3).permute(1, 0, 2, 3).reshape(Co, Ci * kh * kw)
     return _GatherConvFn.apply(xbuf, w1, bias, 1, 1, kh, kw, HO, WO, act)
 
 
