@@ -59,8 +59,25 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        (int)stride);
     return out;
   }
+  // TM=64 is the throughput tile; when its grid would underfill 256 CUs
+  // (< ~3 workgroups/CU) halve the pixel tile to double parallelism — the
+  // small-M resblock convs (40x120) are latency-bound, not FLOP-bound.
+  const long long wgs64 = (long long)grid.x * grid.y * grid.z;
+  if (wgs64 < 256 && M > CONV_TM) {
+    dim3 grid32((M + 31) / 32, grid.y, grid.z);
+    size_t lds32 = (size_t)4 * 32 * (64 + CONV_AP) * 2;
+    hipLaunchKernelGGL(conv_fwd_kernel<32>, grid32, dim3(256), lds32,
+                       at::cuda::getCurrentCUDAStream(),
+                       (const cvbf16*)xbuf.data_ptr(),
+                       (const cvbf16*)wmat.data_ptr(), bptr,
+                       (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
+                       koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
+                       xbuf.stride(0), (long long)N * M, (int)act, (int)WO,
+                       (int)stride);
+    return out;
+  }
   size_t lds = (size_t)4 * CONV_TM * (64 + CONV_AP) * 2;  // 4-buffer pipeline
-  hipLaunchKernelGGL(conv_fwd_kernel, grid, dim3(256), lds,
+  hipLaunchKernelGGL(conv_fwd_kernel<64>, grid, dim3(256), lds,
                      at::cuda::getCurrentCUDAStream(),
                      (const cvbf16*)xbuf.data_ptr(),
                      (const cvbf16*)wmat.data_ptr(), bptr,
@@ -79,8 +96,21 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
   const bool fp8 = xbuf.scalar_type() == torch::kByte;
   const int64_t B = xbuf.size(0);
   const int64_t M = dy.size(2) * dy.size(3);
-  auto dw = torch::zeros({N, K}, xbuf.options().dtype(torch::kFloat32));
-  int pix_chunks = (int)std::min<int64_t>(std::max<int64_t>(M / 2048, 1), 64);
+  // pick pixel chunking so the grid fills the chip (~6 workgroups/CU).
+  // Each (tap-tile, cout-tile, chunk) workgroup owns a disjoint 64x64 region
+  // of its chunk's PARTIAL-SUM slice — plain stores, no atomics (atomicAdd
+  // across ~40 chunks contends on the same dw cache lines and was measured
+  // slower than the 9-chunk version it replaced). The slices are then
+  // reduced with one sum(0). Total partial memory is bounded by the
+  // workgroup target: <= 1536 * 64*64*4B = 25 MB.
+  const int64_t tiles = ((K + 63) / 64) * ((N + 63) / 64) * B;
+  int pix_chunks = (int)std::min<int64_t>(
+      std::max<int64_t>(1536 / std::max<int64_t>(tiles, 1), 1),
+      std::max<int64_t>(M / 256, 1));
+  // every in-range (cout, tap) element of every slice is written exactly
+  // once by its owning workgroup, so empty() needs no zero-fill.
+  auto dwp = torch::empty({(int64_t)B * pix_chunks, N, K},
+                          xbuf.options().dtype(torch::kFloat32));
   dim3 grid((K + 63) / 64, (N + 63) / 64, B * pix_chunks);
   if (fp8) {
     TORCH_CHECK(dy.scalar_type() == torch::kByte, "fp8 wrw: dy must be e4m3");
@@ -88,21 +118,22 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
     hipLaunchKernelGGL(conv_wrw_fp8_kernel, grid, dim3(256), lds,
                        at::cuda::getCurrentCUDAStream(),
                        (const f8*)xbuf.data_ptr(), (const f8*)dy.data_ptr(),
-                       dw.data_ptr<float>(), mbase.data_ptr<int>(),
+                       dwp.data_ptr<float>(), mbase.data_ptr<int>(),
                        koff.data_ptr<int>(), (int)M, (int)N, (int)K,
                        xbuf.stride(0), dy.stride(0), pix_chunks, (int)WO,
                        (int)(mcontig ? 1 : 0));
-    return dw;
+  } else {
+    size_t lds = (size_t)4 * 64 * (32 + CONV_AP) * 2;  // 2 tiles x dbuf
+    hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), lds,
+                       at::cuda::getCurrentCUDAStream(),
+                       (const cvbf16*)xbuf.data_ptr(),
+                       (const cvbf16*)dy.data_ptr(), dwp.data_ptr<float>(),
+                       mbase.data_ptr<int>(), koff.data_ptr<int>(), (int)M,
+                       (int)N, (int)K, xbuf.stride(0), dy.stride(0),
+                       pix_chunks, (int)WO, (int)(mcontig ? 1 : 0));
   }
-  size_t lds = (size_t)4 * 64 * (32 + CONV_AP) * 2;  // 2 tiles x dbuf
-  hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), lds,
-                     at::cuda::getCurrentCUDAStream(),
-                     (const cvbf16*)xbuf.data_ptr(),
-                     (const cvbf16*)dy.data_ptr(), dw.data_ptr<float>(),
-                     mbase.data_ptr<int>(), koff.data_ptr<int>(), (int)M,
-                     (int)N, (int)K, xbuf.stride(0), dy.stride(0), pix_chunks,
-                     (int)WO, (int)(mcontig ? 1 : 0));
-  return dw;
+  if (B * pix_chunks == 1) return dwp.view({N, K});
+  return dwp.sum(0);
 }
 
 }  // namespace dsin

@@ -208,7 +208,7 @@ void conv_wrw_fp8_kernel(const f8* __restrict__ xpad, const f8* __restrict__ dy,
   const int PCHUNK = ((M + pix_chunks - 1) / pix_chunks + 31) & ~31;
   const int p0 = pc * PCHUNK;
   const int p1 = min(p0 + PCHUNK, M);
-  if (p0 >= M) return;
+  // p0 may pass M when PCHUNK rounds up: the tile still stores zeros.
 
   const f8* x = xpad + (long long)img * x_img_stride;
   const f8* g = dy + (long long)img * dy_img_stride;
@@ -255,7 +255,7 @@ void conv_wrw_fp8_kernel(const f8* __restrict__ xpad, const f8* __restrict__ dy,
   write_chunk(0);
   __syncthreads();
 
-  const int nchunks = (p1 - p0 + 31) / 32;
+  const int nchunks = max((p1 - p0 + 31) / 32, 0);
   for (int t = 0; t < nchunks; ++t) {
     if (t + 1 < nchunks) load_chunk(p0 + (t + 1) * 32);
     const char* ac = As + (t & 1) * TBUF;
@@ -282,7 +282,9 @@ void conv_wrw_fp8_kernel(const f8* __restrict__ xpad, const f8* __restrict__ dy,
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int kk = k0 + mi * 16 + kgrp * 4 + reg;
-        if (kk < K) atomicAdd(&dw[(long long)nc * K + kk], acc[mi][reg]);
+        if (kk < K)
+          dw[(long long)blockIdx.z * N * K + (long long)nc * K + kk] =
+              acc[mi][reg];
       }
     }
   }

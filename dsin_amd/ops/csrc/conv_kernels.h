@@ -19,8 +19,10 @@
 // (none/ReLU/leaky-0.2), bf16 store, coalesced within each 16-lane group.
 //
 // conv_wrw_kernel computes dW[co][k] = sum_m dy[co][m] * A[m][k] as a
-// second gather-GEMM (M=K_filter, N=couts, K=pixels) with fp32 atomicAdd
-// accumulation across pixel-chunk workgroups.
+// second gather-GEMM (M=K_filter, N=couts, K=pixels); each pixel-chunk
+// workgroup stores its 64x64 tile into its own partial-sum slice of
+// dw[B*pix_chunks][N][K] (plain stores — no atomics) and the host reduces
+// the slices with one sum(0).
 
 #pragma once
 #include <hip/hip_runtime.h>
@@ -59,6 +61,12 @@ __global__ void conv_tables_kernel(int* __restrict__ mbase,
   }
 }
 
+// TM = output pixels per workgroup. 64 is the throughput tile; 32 doubles
+// the workgroup count for small-M layers (e.g. the 40x120 resblock convs,
+// which at TM=64 launch only ~600 workgroups on 256 CUs and run
+// latency-bound at ~2.3 waves/SIMD). The host picks 32 when the TM=64 grid
+// would underfill the chip.
+template <int TM>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
                      const cvbf16* __restrict__ wmat,   // (Co, KP64+AP) 0-pad
@@ -80,16 +88,20 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   // of the A loads so the compiler's wait before the MFMAs is a counted
   // vmcnt, not a pipeline-draining vmcnt(0).
   const int KC = 64;
+  constexpr int MI = TM / 16;       // MFMA row-subtiles per wave
+  constexpr int PPT = TM / 8;       // pixel octets per k column (staging)
+  constexpr int KCOV = 256 / PPT;   // k columns covered per staging pass
+  constexpr int NST = 64 / KCOV;    // staging passes per 64-chunk (2 or 1)
   const int WSTRIDE = KP + CONV_AP;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int ASTR = KC + CONV_AP;
-  const int ABUF = CONV_TM * ASTR;
-  cvbf16* As = reinterpret_cast<cvbf16*>(smem);          // 2 x [TM][KC+AP]
+  const int ABUF = TM * ASTR;
+  cvbf16* As = reinterpret_cast<cvbf16*>(smem);          // 4 x [TM][KC+AP]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int m0 = blockIdx.x * CONV_TM;
+  const int m0 = blockIdx.x * TM;
   const int n0 = blockIdx.y * CONV_TN + wid * 16;
   const long long img = blockIdx.z;
   const cvbf16* x = xpad + img * x_img_stride;
@@ -97,12 +109,13 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   const int colL = lane & 15;
   const int kgrp = lane >> 4;
 
-  cv_f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                     {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  cv_f32x4 acc[MI];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) acc[mi] = {0.f, 0.f, 0.f, 0.f};
 
-  // staging: thread t -> 8 consecutive output pixels at k = (t>>3), k+32
-  const int sm8 = (tid & 7) * 8;
-  const int sk = tid >> 3;                 // 0..31
+  // staging: thread t -> 8 consecutive output pixels at k-column t / PPT
+  const int sm8 = (tid % PPT) * 8;
+  const int sk = tid / PPT;                // 0..KCOV-1
   const int gm0 = min(m0 + sm8, M - 1);
   const int gm7 = min(m0 + sm8 + 7, M - 1);
   const int mb0 = mbase[gm0];
@@ -116,7 +129,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   const cvbf16* wrow = wmat + (long long)(ncol < N ? ncol : 0) * WSTRIDE;
 
   typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
-  u16x8 stage[2];
+  u16x8 stage[NST];
   cv_bf16x8 wfrag[2];
 
   auto load_half = [&](int k, u16x8& st) {
@@ -153,8 +166,8 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
     // vmcnt that leaves the later A loads in flight
     wfrag[0] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + kgrp * 8]);
     wfrag[1] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + 32 + kgrp * 8]);
-    load_half(kc + sk, stage[0]);
-    load_half(kc + 32 + sk, stage[1]);
+#pragma unroll
+    for (int h = 0; h < NST; ++h) load_half(kc + h * KCOV + sk, stage[h]);
   };
   // A-tile byte-address XOR swizzle (see write/read pair): staging writes at
   // an 8-row stride collide on banks; rows stay 16B aligned and 128B blocks
@@ -165,22 +178,22 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   char* As8 = reinterpret_cast<char*>(As);
   // chunk-invariant offsets, computed once (the staging/read address math
   // was ~12 VALU per MFMA when recomputed per chunk)
-  int wr_off[16];
+  int wr_off[NST * 8];
 #pragma unroll
-  for (int h = 0; h < 2; ++h)
+  for (int h = 0; h < NST; ++h)
 #pragma unroll
     for (int i = 0; i < 8; ++i)
-      wr_off[h * 8 + i] = aswz(sm8 + i, h * 32 + sk);
-  int rd_off[8];
+      wr_off[h * 8 + i] = aswz(sm8 + i, h * KCOV + sk);
+  int rd_off[2 * MI];
 #pragma unroll
   for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-      rd_off[kk * 4 + mi] = aswz(mi * 16 + colL, kk * 32 + kgrp * 8);
+    for (int mi = 0; mi < MI; ++mi)
+      rd_off[kk * MI + mi] = aswz(mi * 16 + colL, kk * 32 + kgrp * 8);
   auto write_chunk = [&](int buf) {
     char* dst = As8 + buf * ABUF * 2;
 #pragma unroll
-    for (int h = 0; h < 2; ++h)
+    for (int h = 0; h < NST; ++h)
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         unsigned short u = stage[h][i];
@@ -195,18 +208,18 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   // chunk kt, writes S1 into buf[(kt+1)&3], barriers; odd iterations swap
   // the sets. A-prefetch therefore has ~two MFMA phases to land.
   const int nchunks = KP / KC;
-  u16x8 stageB[2];
+  u16x8 stageB[NST];
   cv_bf16x8 wfragB[2];
   auto load_chunkB = [&](int kc) {
     wfragB[0] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + kgrp * 8]);
     wfragB[1] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + 32 + kgrp * 8]);
-    load_half(kc + sk, stageB[0]);
-    load_half(kc + 32 + sk, stageB[1]);
+#pragma unroll
+    for (int h = 0; h < NST; ++h) load_half(kc + h * KCOV + sk, stageB[h]);
   };
   auto write_chunkB = [&](int buf) {
     char* dst = As8 + buf * ABUF * 2;
 #pragma unroll
-    for (int h = 0; h < 2; ++h)
+    for (int h = 0; h < NST; ++h)
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         unsigned short u = stageB[h][i];
@@ -219,9 +232,9 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
+      for (int mi = 0; mi < MI; ++mi) {
         const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
-            &cur[rd_off[kk * 4 + mi]]);
+            &cur[rd_off[kk * MI + mi]]);
         acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, w2[kk],
                                                           acc[mi], 0, 0, 0);
       }
@@ -267,7 +280,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   cvbf16* o = out + img * o_img_stride + (long long)(ncol < N ? ncol : 0) * M;
   if (ncol < N) {
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
+    for (int mi = 0; mi < MI; ++mi) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int m = m0 + mi * 16 + kgrp * 4 + reg;
@@ -288,7 +301,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 __global__ __launch_bounds__(256)
 void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
                      const cvbf16* __restrict__ dy,    // (Co, M)
-                     float* __restrict__ dw,           // (Co, K) fp32 accum
+                     float* __restrict__ dw,           // (B*chunks, Co, K)
                      const int* __restrict__ mbase,
                      const int* __restrict__ koff,
                      int M, int N, int K,
@@ -310,7 +323,8 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
   const int PCHUNK = ((M + pix_chunks - 1) / pix_chunks + 31) & ~31;
   const int p0 = pc * PCHUNK;
   const int p1 = min(p0 + PCHUNK, M);
-  if (p0 >= M) return;
+  // p0 may pass M when PCHUNK's 32-alignment rounds up: the tile still
+  // stores (zeros) — its slice is reduced by the host's sum(0).
 
   const cvbf16* x = xpad + (long long)img * x_img_stride;
   const cvbf16* g = dy + (long long)img * dy_img_stride;
@@ -367,7 +381,7 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
   write_chunk(0);
   __syncthreads();
 
-  const int nchunks = (p1 - p0 + 31) / 32;
+  const int nchunks = max((p1 - p0 + 31) / 32, 0);
   for (int t = 0; t < nchunks; ++t) {
     if (t + 1 < nchunks) load_chunk(p0 + (t + 1) * 32);
     const cvbf16* ac = As + (t & 1) * TBUF;
@@ -387,7 +401,8 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
     }
   }
 
-  // D[row=tap][col=cout]; accumulate into dw[cout][tap]
+  // D[row=tap][col=cout]; store into this workgroup's slice of dw
+  float* dws = dw + (long long)blockIdx.z * N * K;
   const int nc = blockIdx.y * 64 + wid * 16 + colL;
   if (nc < N) {
 #pragma unroll
@@ -395,7 +410,7 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int kk = k0 + mi * 16 + kgrp * 4 + reg;
-        if (kk < K) atomicAdd(&dw[(long long)nc * K + kk], acc[mi][reg]);
+        if (kk < K) dws[(long long)nc * K + kk] = acc[mi][reg];
       }
     }
   }

@@ -11,6 +11,10 @@
 
 namespace dsin {
 
+// one block row-group per (b, c, output row); threads stride the row.
+// Row-oriented addressing keeps the per-element index math 32-bit and
+// branch-uniform (the old flat-index version spent most of its time in
+// 64-bit div/mod per element: 13 us for a 4 MB buffer).
 template <typename T>
 __global__ void pad_stuff_kernel(const T* __restrict__ x,
                                  bf16* __restrict__ out,
@@ -18,31 +22,24 @@ __global__ void pad_stuff_kernel(const T* __restrict__ x,
                                  int pt, int pl, int stride,
                                  long long n_img_out, long long n_img_in,
                                  int B) {
-  long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  long long total = (long long)B * n_img_out;
-  long long gstride = (long long)gridDim.x * blockDim.x;
-  for (; idx < total; idx += gstride) {
-    long long b = idx / n_img_out;
-    long long rem = idx % n_img_out;
-    int c = rem / (Hp * Wp);
-    int r2 = rem % (Hp * Wp);
-    int i = r2 / Wp, j = r2 % Wp;
+  const int row = blockIdx.x;          // (b*C + c)*Hp + i
+  const int i = row % Hp;
+  const int bc = row / Hp;
+  bf16* orow = out + (long long)bc * Hp * Wp + (long long)i * Wp;
+  const int ii = i - pt;
+  const bool rowin = ii >= 0 && (stride == 1 || ii % stride == 0) &&
+                     ii / stride < H;
+  if (!rowin) {
+    for (int j = threadIdx.x; j < Wp; j += blockDim.x) orow[j] = f2b(0.f);
+    return;
+  }
+  const T* xrow = x + ((long long)bc * H + ii / stride) * W;
+  for (int j = threadIdx.x; j < Wp; j += blockDim.x) {
+    const int jj = j - pl;
     float v = 0.f;
-    int ii = i - pt, jj = j - pl;
-    if (ii >= 0 && jj >= 0) {
-      if (stride == 1) {
-        if (ii < H && jj < W)
-          v = (float)x[b * n_img_in + ((long long)c * H + ii) * W + jj];
-      } else {
-        if (ii % stride == 0 && jj % stride == 0) {
-          ii /= stride;
-          jj /= stride;
-          if (ii < H && jj < W)
-            v = (float)x[b * n_img_in + ((long long)c * H + ii) * W + jj];
-        }
-      }
-    }
-    out[idx] = f2b(v);
+    if (jj >= 0 && (stride == 1 || jj % stride == 0) && jj / stride < W)
+      v = (float)xrow[jj / stride];
+    orow[j] = f2b(v);
   }
 }
 
@@ -61,6 +58,8 @@ torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
   auto out = store.narrow(0, 0, B * n_img).view({B, C, Hp, Wp});
   long long total = (long long)B * n_img;
   int grid = (int)std::min<long long>((total + 255) / 256, 8192);
+  const int rows = B * C * Hp;
+  const int rthreads = Wp >= 256 ? 256 : (Wp >= 128 ? 128 : 64);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   if (fp8) {
     if (x.scalar_type() == torch::kFloat32) {
@@ -79,13 +78,13 @@ torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
     return out;
   }
   if (x.scalar_type() == torch::kFloat32) {
-    hipLaunchKernelGGL((pad_stuff_kernel<float>), dim3(grid), dim3(256), 0,
-                       stream, x.data_ptr<float>(), (bf16*)out.data_ptr(), C,
-                       H, W, Hp, Wp, (int)pt, (int)pl, (int)stride, n_img,
+    hipLaunchKernelGGL((pad_stuff_kernel<float>), dim3(rows), dim3(rthreads),
+                       0, stream, x.data_ptr<float>(), (bf16*)out.data_ptr(),
+                       C, H, W, Hp, Wp, (int)pt, (int)pl, (int)stride, n_img,
                        (long long)C * H * W, B);
   } else if (x.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL((pad_stuff_kernel<bf16>), dim3(grid), dim3(256), 0,
-                       stream, (const bf16*)x.data_ptr(),
+    hipLaunchKernelGGL((pad_stuff_kernel<bf16>), dim3(rows), dim3(rthreads),
+                       0, stream, (const bf16*)x.data_ptr(),
                        (bf16*)out.data_ptr(), C, H, W, Hp, Wp, (int)pt,
                        (int)pl, (int)stride, n_img, (long long)C * H * W, B);
   } else {
