@@ -63,14 +63,34 @@ def _wmat(w1: torch.Tensor, fp8: bool = False) -> torch.Tensor:
     """(Co, K) any-dtype -> bf16 (or e4m3-as-uint8) zero-padded to
     (Co, KP64+8); the zeros cancel the clamped out-of-range A gathers."""
     K = w1.shape[1]
-    KP = (K + 63) & ~63
     if fp8:
+        KP = (K + 63) & ~63
         w8 = w1.to(torch.float8_e4m3fn).view(torch.uint8)
         out = torch.zeros(w1.shape[0], KP + 8, dtype=torch.uint8,
                           device=w1.device)
         out[:, :K] = w8
         return out
+    if w1.is_cuda and hip_available():
+        return _require_ext("wmat_make")(w1.contiguous(), 1, False)
+    KP = (K + 63) & ~63
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
+
+
+def _wmat_rot(w1: torch.Tensor, co: int, ci: int, khw: int) -> torch.Tensor:
+    """Padded bf16 W panel for the backward-data gather conv: spatial taps
+    reversed and cin/cout swapped — one kernel instead of the torch
+    flip+permute+reshape+pad chain (~4 kernels per conv backward)."""
+    return _require_ext("wmat_make")(w1.contiguous(), khw, True)
+
+
+def _act_grad(dy: torch.Tensor, y_act, act: int) -> torch.Tensor:
+    """dy * act'(y) fused to one kernel, bf16 out (act=0: plain cast)."""
+    dy = dy.contiguous()
+    if act:
+        return _require_ext("act_bwd")(dy, y_act, act)
+    if dy.dtype != torch.bfloat16:
+        dy = dy.to(torch.bfloat16)
+    return dy
 
 
 class _PadStuffFn(torch.autograd.Function):
@@ -140,23 +160,16 @@ class _GatherConvFn(torch.autograd.Function):
         B, Ci, Hp, Wp = xbuf.shape
         Co, K = w1.shape
 
-        dy = dy.contiguous().to(torch.bfloat16)
-        if act == 1:
-            dy = torch.where(y_act > 0, dy, torch.zeros((), dtype=dy.dtype,
-                                                        device=dy.device))
-        elif act == 2:
-            dy = torch.where(y_act > 0, dy, dy * 0.2)
+        dy = _act_grad(dy, y_act, act)
         dxbuf = None
         if ctx.needs_input_grad[0]:
             with torch.no_grad():
                 pe_h, pe_w = (kh - 1) * dil, (kw - 1) * dil
                 dybuf = _padded_buf(dy, pe_w, pe_w, pe_h, pe_h, stride=stride)
-                wrot = (w1.view(Co, Ci, kh, kw).flip(2, 3).permute(1, 0, 2, 3)
-                        .reshape(Ci, Co * kh * kw))
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
                                  kh, kw, 1, dil, Hp, Wp)
-                dxbuf = ext_fwd(dybuf, _wmat(wrot), None, mb2, ko2, Ci,
-                                Co * kh * kw, Hp, Wp, 0, 1)
+                dxbuf = ext_fwd(dybuf, _wmat_rot(w1, Co, Ci, kh * kw), None,
+                                mb2, ko2, Ci, Co * kh * kw, Hp, Wp, 0, 1)
 
         dw1 = None
         if ctx.needs_input_grad[1]:
@@ -360,25 +373,20 @@ class _GatherConv3dFn(torch.autograd.Function):
         ext_wrw = _require_ext("conv_wrw")
         B, Ci, Dp, Hp, Wp = xbuf.shape
         Co, K = w1.shape
-        dy = dy.contiguous().to(torch.bfloat16)
-        if act == 1:
-            dy = torch.where(y_act.view_as(dy) > 0, dy,
-                             torch.zeros((), dtype=dy.dtype, device=dy.device))
+        dy = _act_grad(dy.contiguous(), y_act, act).view(B, Co, Do, Ho, Wo)
 
         dxbuf = None
         if ctx.needs_input_grad[0]:
             with torch.no_grad():
                 dybuf = _padded_buf3d(dy, kd - 1, kh - 1, kw - 1)
-                wrot = (w1.view(Co, Ci, kd, kh, kw).flip(2, 3, 4)
-                        .permute(1, 0, 2, 3, 4).reshape(Ci, Co * kd * kh * kw))
                 mb2, ko2, D2, H2, W2 = _plan3d(dy.device, Co, dybuf.shape[2],
                                                dybuf.shape[3], dybuf.shape[4],
                                                kd, kh, kw)
                 assert (D2, H2, W2) == (Dp, Hp, Wp)
                 dxbuf = ext_fwd(dybuf.view(B, Co, -1, dybuf.shape[4]),
-                                _wmat(wrot), None, mb2, ko2, Ci,
-                                Co * kd * kh * kw, Dp * Hp, Wp, 0,
-                                1).view(B, Ci, Dp, Hp, Wp)
+                                _wmat_rot(w1, Co, Ci, kd * kh * kw), None,
+                                mb2, ko2, Ci, Co * kd * kh * kw, Dp * Hp, Wp,
+                                0, 1).view(B, Ci, Dp, Hp, Wp)
 
         dw1 = None
         if ctx.needs_input_grad[1]:

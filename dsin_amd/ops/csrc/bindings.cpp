@@ -27,6 +27,8 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
 torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t WO, bool mcontig);
+torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act);
+torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, bool rot);
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor lr, torch::Tensor step,
                double b1, double b2, double eps);
@@ -54,6 +56,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_tables", &dsin::conv_tables, "gather-conv offset tables");
   m.def("conv_fwd", &dsin::conv_fwd, "implicit-GEMM gather conv forward");
   m.def("conv_wrw", &dsin::conv_wrw, "implicit-GEMM conv weight gradient");
+  m.def("act_bwd", &dsin::act_bwd, "fused activation gradient (bf16 out)");
+  m.def("wmat_make", &dsin::wmat_make, "padded/rotated bf16 conv W panel");
   m.def("adam_step", &dsin::adam_step, "fused flat-buffer Adam step");
   m.def("pad_stuff", &dsin::pad_stuff, "fused pad/zero-stuff/cast to bf16");
   m.def("bn_fwd", &dsin::bn_fwd, "fused batch-norm(+act) forward");

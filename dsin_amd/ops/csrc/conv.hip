@@ -136,4 +136,52 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
   return dwp.sum(0);
 }
 
+torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act) {
+  CHECK_CUDA_CONTIG(dy);
+  CHECK_CUDA_CONTIG(y);
+  const long long n = dy.numel();
+  auto out = torch::empty_like(y);  // y is bf16, same shape
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int grid = (int)std::min<long long>((n + 255) / 256, 4096);
+  if (dy.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((act_bwd_kernel<float>), dim3(grid), dim3(256), 0,
+                       stream, dy.data_ptr<float>(),
+                       (const cvbf16*)y.data_ptr(), (cvbf16*)out.data_ptr(), n,
+                       (int)act);
+  } else {
+    TORCH_CHECK(dy.scalar_type() == torch::kBFloat16, "act_bwd: fp32/bf16");
+    hipLaunchKernelGGL((act_bwd_kernel<cvbf16>), dim3(grid), dim3(256), 0,
+                       stream, (const cvbf16*)dy.data_ptr(),
+                       (const cvbf16*)y.data_ptr(), (cvbf16*)out.data_ptr(), n,
+                       (int)act);
+  }
+  return out;
+}
+
+torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, bool rot) {
+  CHECK_CUDA_CONTIG(w1);
+  const int64_t N = w1.size(0), K = w1.size(1);
+  const int64_t rows = rot ? K / khw : N;
+  const int64_t kout = rot ? N * khw : K;
+  const int64_t KPA = ((kout + 63) & ~63) + CONV_AP;
+  auto out = torch::empty({rows, KPA},
+                          w1.options().dtype(torch::kBFloat16));
+  const long long total = rows * KPA;
+  const int grid = (int)std::min<long long>((total + 255) / 256, 4096);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (w1.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((wmat_make_kernel<float>), dim3(grid), dim3(256), 0,
+                       stream, w1.data_ptr<float>(), (cvbf16*)out.data_ptr(),
+                       (int)rows, (int)kout, (int)K, (int)khw, (int)KPA,
+                       rot ? 1 : 0);
+  } else {
+    TORCH_CHECK(w1.scalar_type() == torch::kBFloat16, "wmat_make: fp32/bf16");
+    hipLaunchKernelGGL((wmat_make_kernel<cvbf16>), dim3(grid), dim3(256), 0,
+                       stream, (const cvbf16*)w1.data_ptr(),
+                       (cvbf16*)out.data_ptr(), (int)rows, (int)kout, (int)K,
+                       (int)khw, (int)KPA, rot ? 1 : 0);
+  }
+  return out;
+}
+
 }  // namespace dsin

@@ -132,8 +132,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   u16x8 stage[NST];
   cv_bf16x8 wfrag[2];
 
-  auto load_half = [&](int k, u16x8& st) {
-    const int ko = (k < K) ? koff[k] : 0;
+  auto load_half = [&](int ko, u16x8& st) {
     if (stride == 1) {
       const u16x8 a = *reinterpret_cast<const u16x8*>(&x[mb0 + ko]);
       const u16x8 b = *reinterpret_cast<const u16x8*>(
@@ -161,13 +160,26 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
       }
     }
   };
+  // koff values are PREFETCHED two chunks ahead into registers: loading
+  // koff[k] at staging time serializes the chain koff -> wait -> A-load and
+  // was the dominant stall (each set stages every other chunk, so its next
+  // k offsets are +2*KC ahead).
+  auto ko_at = [&](int k) { return (k < K) ? koff[k] : 0; };
+  int koA[NST], koB[NST];
+#pragma unroll
+  for (int h = 0; h < NST; ++h) {
+    koA[h] = ko_at(h * KCOV + sk);
+    koB[h] = ko_at(KC + h * KCOV + sk);
+  }
   auto load_chunk = [&](int kc) {
     // W first: its consumer (the MFMA phase) waits on it with a counted
     // vmcnt that leaves the later A loads in flight
     wfrag[0] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + kgrp * 8]);
     wfrag[1] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + 32 + kgrp * 8]);
 #pragma unroll
-    for (int h = 0; h < NST; ++h) load_half(kc + h * KCOV + sk, stage[h]);
+    for (int h = 0; h < NST; ++h) load_half(koA[h], stage[h]);
+#pragma unroll
+    for (int h = 0; h < NST; ++h) koA[h] = ko_at(kc + 2 * KC + h * KCOV + sk);
   };
   // A-tile byte-address XOR swizzle (see write/read pair): staging writes at
   // an 8-row stride collide on banks; rows stay 16B aligned and 128B blocks
@@ -214,7 +226,9 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
     wfragB[0] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + kgrp * 8]);
     wfragB[1] = *reinterpret_cast<const cv_bf16x8*>(&wrow[kc + 32 + kgrp * 8]);
 #pragma unroll
-    for (int h = 0; h < NST; ++h) load_half(kc + h * KCOV + sk, stageB[h]);
+    for (int h = 0; h < NST; ++h) load_half(koB[h], stageB[h]);
+#pragma unroll
+    for (int h = 0; h < NST; ++h) koB[h] = ko_at(kc + 2 * KC + h * KCOV + sk);
   };
   auto write_chunkB = [&](int buf) {
     char* dst = As8 + buf * ABUF * 2;
@@ -418,18 +432,51 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
 
 // activation-gradient helper for fused lrelu/relu epilogues:
 // dyp = dy * act'(y) computed from the post-activation output y.
-__global__ void act_bwd_kernel(const cvbf16* __restrict__ dy,
+// Replaces the torch chain fill-scalar + where + cast (3 kernels) per
+// conv backward with one kernel; dy may be fp32 or bf16.
+template <typename T>
+__global__ void act_bwd_kernel(const T* __restrict__ dy,
                                const cvbf16* __restrict__ y,
                                cvbf16* __restrict__ out,
                                long long n, int act) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   long long stride = (long long)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
-    float g = cvb2f(dy[i]);
+    float g = (float)dy[i];
     float yy = cvb2f(y[i]);
     if (act == 1) g = yy > 0.f ? g : 0.f;
     else if (act == 2) g = yy > 0.f ? g : 0.2f * g;
     out[i] = cvf2b(g);
+  }
+}
+
+// builds the zero-padded bf16 W panel for conv_fwd in ONE kernel, replacing
+// the torch pad+contiguous (forward) or flip+permute+reshape+pad chain
+// (backward-data "rotated" weights, ~4 kernels per conv backward).
+//   rot=0: out[n][k] = w1[n][k]                       (N=Co rows, K taps)
+//   rot=1: out[ci][co*khw + t] = w1[co][ci*khw + khw-1-t]
+// (full multi-radix tap reversal == flip of every spatial dim), k >= K -> 0.
+template <typename T>
+__global__ void wmat_make_kernel(const T* __restrict__ w1,
+                                 cvbf16* __restrict__ out,
+                                 int rows, int kout, int kin, int khw,
+                                 int KPA, int rot) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = (long long)rows * KPA;
+  long long gstride = (long long)gridDim.x * blockDim.x;
+  for (; i < total; i += gstride) {
+    const int n = (int)(i / KPA);
+    const int k = (int)(i % KPA);
+    float v = 0.f;
+    if (k < kout) {
+      if (rot) {
+        const int co = k / khw, t = k % khw;
+        v = (float)w1[(long long)co * kin + n * khw + (khw - 1 - t)];
+      } else {
+        v = (float)w1[(long long)n * kin + k];
+      }
+    }
+    out[i] = cvf2b(v);
   }
 }
 
