@@ -93,7 +93,13 @@ class DSIN(nn.Module):
         weights, the centers term, plus PC when enabled (reference
         src/Distortions_imgcomp.py:129-136, src/quantizer_imgcomp.py:18-24).
         Computed as ONE flattened reduction (a per-tensor loop costs ~400
-        kernel launches per step)."""
+        kernel launches per step). When a Trainer has folded the L2 terms
+        into the fused optimizer (training/trainer.py), reg_value_fn
+        reports the identical value from the optimizer's flat buffers and
+        the autograd subgraph is skipped entirely."""
+        fn = getattr(self, "reg_value_fn", None)
+        if fn is not None:
+            return fn()
         dev = self.encoder.quantizer.centers.device
         s = torch.zeros((), device=dev)
         if self.reg_factor:

@@ -50,6 +50,7 @@ class FusedAdam:
 
         self.lr_t = torch.tensor(float(lr), device=device)
         self.step_t = torch.zeros(1, device=device, dtype=torch.int32)
+        self.wd = None  # optional per-element L2 factors (see set_weight_decay)
         self.betas = tuple(betas)
         self.eps = float(eps)
         # torch-optimizer-compatible surface for the LR schedule
@@ -60,12 +61,33 @@ class FusedAdam:
         # set_to_none would break the flat-buffer invariant; always zero
         self.flat_g.zero_()
 
+    def set_weight_decay(self, param: torch.nn.Parameter, factor: float) -> bool:
+        """Register a classic L2 term factor/2*sum(p^2) for `param`: its
+        lambda*p gradient is added inside the fused step (identical Adam
+        update to building the regularizer in the autograd graph, without
+        the ~100 cat/split/add kernels per step that graph costs)."""
+        for p, (ofs, n) in zip(self.params, self._slices):
+            if p is param:
+                if self.wd is None:
+                    self.wd = torch.zeros_like(self.flat_g)
+                self.wd[ofs:ofs + n].fill_(float(factor))
+                return True
+        return False
+
+    @torch.no_grad()
+    def reg_value(self) -> torch.Tensor:
+        """Current value of all registered L2 terms: 0.5*sum(wd * p^2)."""
+        if self.wd is None:
+            return torch.zeros((), device=self.flat_p.device)
+        return 0.5 * (self.wd * self.flat_p.square()).sum()
+
     @torch.no_grad()
     def step(self):
         step_fn = _require_ext("adam_step")
         self.step_t += 1
         step_fn(self.flat_p, self.flat_g, self.exp_avg, self.exp_avg_sq,
-                self.lr_t, self.step_t, self.betas[0], self.betas[1], self.eps)
+                self.lr_t, self.step_t, self.wd, self.betas[0], self.betas[1],
+                self.eps)
 
     def state_dict(self):
         return {

@@ -178,7 +178,8 @@ class _GatherConvFn(torch.autograd.Function):
             dw1 = ext_wrw(xbuf, dy, mbase, koff, Co, K, WO,
                           stride == 1).to(w1.dtype)
 
-        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        dbias = (dy.sum(dim=(0, 2, 3), dtype=torch.float32)
+                 if has_bias else None)
         return dxbuf, dw1, dbias, None, None, None, None, None, None, None
 
 
@@ -248,7 +249,8 @@ class _GatherConvFP8Fn(torch.autograd.Function):
             dw1 = ext_wrw(xbuf, dy8, mbase, koff, Co, K, WO,
                           stride == 1).to(w1.dtype)
 
-        dbias = dy.float().sum(dim=(0, 2, 3)) if has_bias else None
+        dbias = (dy.sum(dim=(0, 2, 3), dtype=torch.float32)
+                 if has_bias else None)
         return (dx, dw1, dbias, None, None, None, None, None, None, None,
                 None, None)
 
@@ -396,7 +398,8 @@ class _GatherConv3dFn(torch.autograd.Function):
                           dy.view(B, Co, Do * Ho, Wo), mbase, koff, Co, K,
                           Wo, True).to(w1.dtype)
 
-        dbias = dy.float().sum(dim=(0, 2, 3, 4)) if has_bias else None
+        dbias = (dy.sum(dim=(0, 2, 3, 4), dtype=torch.float32)
+                 if has_bias else None)
         return dxbuf, dw1, dbias, None, None, None, None
 
 

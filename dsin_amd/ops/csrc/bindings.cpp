@@ -31,7 +31,8 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act);
 torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, bool rot);
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor lr, torch::Tensor step,
-               double b1, double b2, double eps);
+               c10::optional<torch::Tensor> wd, double b1, double b2,
+               double eps);
 torch::Tensor pad_stuff(torch::Tensor x, int64_t pt, int64_t pb, int64_t pl,
                         int64_t pr, int64_t stride, bool fp8);
 std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,

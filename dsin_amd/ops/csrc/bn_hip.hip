@@ -45,8 +45,8 @@ __device__ __forceinline__ void block_reduce2(float& a, float& b) {
 }
 
 __global__ void bn_stats_part_kernel(const bnbf16* __restrict__ y,
-                                     float* __restrict__ psum,   // (C,)
-                                     float* __restrict__ psum2,  // (C,)
+                                     float* __restrict__ psum,   // (S, C)
+                                     float* __restrict__ psum2,  // (S, C)
                                      int C, long long HW, int B, int S) {
   const int c = blockIdx.y;
   const int sidx = blockIdx.x;
@@ -71,13 +71,16 @@ __global__ void bn_stats_part_kernel(const bnbf16* __restrict__ y,
     }
   }
   block_reduce2(s, s2);
+  // per-slice plain stores into (S, C) partials: no atomics, and the
+  // buffers need no zero-fill kernel (two fills per BN call was ~0.6 ms
+  // per training step); bn_finalize sums the S slices.
   if (threadIdx.x == 0) {
-    atomicAdd(&psum[c], s);
-    atomicAdd(&psum2[c], s2);
+    psum[sidx * C + c] = s;
+    psum2[sidx * C + c] = s2;
   }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ psum,
+__global__ void bn_finalize_kernel(const float* __restrict__ psum,   // (S,C)
                                    const float* __restrict__ psum2,
                                    float* __restrict__ mean,
                                    float* __restrict__ rstd,
@@ -87,14 +90,19 @@ __global__ void bn_finalize_kernel(const float* __restrict__ psum,
                                    const float* __restrict__ beta,
                                    float* __restrict__ rmean,
                                    float* __restrict__ rvar,
-                                   int C, float n, float momentum, float eps,
-                                   int training) {
+                                   int C, int S, float n, float momentum,
+                                   float eps, int training) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float m, rs;
   if (training) {
-    m = psum[c] / n;
-    float var = fmaxf(psum2[c] / n - m * m, 0.f);
+    float t1 = 0.f, t2 = 0.f;
+    for (int sdx = 0; sdx < S; ++sdx) {
+      t1 += psum[sdx * C + c];
+      t2 += psum2[sdx * C + c];
+    }
+    m = t1 / n;
+    float var = fmaxf(t2 / n - m * m, 0.f);
     rs = rsqrtf(var + eps);
     const float unbias = (n > 1.f) ? var * n / (n - 1.f) : var;
     rmean[c] = (1.f - momentum) * rmean[c] + momentum * m;
@@ -272,18 +280,18 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
   auto shift = torch::empty({C}, optsF);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   if (training) {
-    auto psum = torch::zeros({C}, optsF);
-    auto psum2 = torch::zeros({C}, optsF);
     const int S = _spatial_chunks(HW);
+    auto parts = torch::empty({2, S, C}, optsF);  // written fully: no fill
+    float* psum = parts.data_ptr<float>();
+    float* psum2 = psum + (long long)S * C;
     hipLaunchKernelGGL(bn_stats_part_kernel, dim3(S, C), dim3(256), 0, stream,
-                       (const bnbf16*)y.data_ptr(), psum.data_ptr<float>(),
-                       psum2.data_ptr<float>(), C, HW, B, S);
+                       (const bnbf16*)y.data_ptr(), psum, psum2, C, HW, B, S);
     hipLaunchKernelGGL(bn_finalize_kernel, grid1d(C, 256), dim3(256), 0,
-                       stream, psum.data_ptr<float>(), psum2.data_ptr<float>(),
+                       stream, psum, psum2,
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
                        scale.data_ptr<float>(), shift.data_ptr<float>(),
                        gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                       rmean.data_ptr<float>(), rvar.data_ptr<float>(), C,
+                       rmean.data_ptr<float>(), rvar.data_ptr<float>(), C, S,
                        (float)((long long)B * HW), (float)momentum, (float)eps,
                        1);
   } else {
@@ -292,7 +300,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
                        rstd.data_ptr<float>(), scale.data_ptr<float>(),
                        shift.data_ptr<float>(), gamma.data_ptr<float>(),
                        beta.data_ptr<float>(), rmean.data_ptr<float>(),
-                       rvar.data_ptr<float>(), C, (float)((long long)B * HW),
+                       rvar.data_ptr<float>(), C, 0,
+                       (float)((long long)B * HW),
                        (float)momentum, (float)eps, 0);
   }
   auto out = torch::empty_like(y);
@@ -318,8 +327,9 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   const int B = (int)y.size(0), C = (int)y.size(1);
   const long long HW = (long long)y.size(2) * y.size(3);
   auto optsF = y.options().dtype(torch::kFloat32);
-  auto s1 = torch::zeros({C}, optsF);
-  auto s2 = torch::zeros({C}, optsF);
+  auto sbuf = torch::zeros({2, C}, optsF);  // one fill for both sums
+  auto s1 = sbuf[0];
+  auto s2 = sbuf[1];
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int S = _spatial_chunks(HW);
   hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(S, C), dim3(256), 0, stream,

@@ -94,6 +94,7 @@ class Trainer:
         from ..ops.adam import FusedAdam
         self._fused = isinstance(self.opt_ae, FusedAdam)
         if self._fused:
+            self._delegate_reg_to_optimizer(model)
             # flat gradient buffers double as the DDP communication buffers
             self.reducer = GradReducer([])  # inactive
         else:
@@ -124,6 +125,33 @@ class Trainer:
             return torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                                   cache_enabled=not self.use_cuda_graph)
         return contextlib.nullcontext()
+
+
+    def _delegate_reg_to_optimizer(self, model) -> None:
+        """Fold the model's factor/2*sum(w^2) regularizers into the fused
+        Adam step as per-element L2 factors (mathematically identical
+        update; see ops/adam.py). The loss-side value is then computed from
+        the optimizer's flat buffer instead of a ~100-kernel autograd
+        subgraph per step."""
+        import torch.nn as nn
+        ok = True
+        if getattr(model, "reg_factor", 0.0):
+            for mod in (model.encoder, model.decoder):
+                for m in mod.modules():
+                    if isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
+                        ok &= self.opt_ae.set_weight_decay(
+                            m.weight, model.reg_factor)
+        q = model.encoder.quantizer
+        if q.reg_factor:
+            ok &= self.opt_ae.set_weight_decay(q.centers, q.reg_factor)
+        pc = model.probclass
+        if pc.reg_factor is not None:
+            for mod in (pc.conv0, pc.res_conv1, pc.res_conv2, pc.conv2):
+                ok &= self.opt_pc.set_weight_decay(mod.weight,
+                                                   float(pc.reg_factor))
+        if ok:
+            model.reg_value_fn = lambda: (self.opt_ae.reg_value()
+                                          + self.opt_pc.reg_value())
 
     def _step_inner(self, x: torch.Tensor, y: Optional[torch.Tensor]):
         """zero-grad + forward + backward + both optimizer steps. This is
