@@ -35,6 +35,13 @@ def build_configs(args):
     ae, _ = config_mod.parse(os.path.join(cur, "run_configs", "ae_run_configs"))
     pc, _ = config_mod.parse(os.path.join(cur, "run_configs", "pc_run_configs"))
     ae.crop_size = (args.height, args.width)
+    # keep the siFinder patch grid valid for non-default crops (the
+    # reference's (20, 24) patches assume 320x960 / 320x1224 crops)
+    ph, pw = ae.y_patch_size
+    if args.height % ph or args.width % pw:
+        ph = next(p for p in (20, 16, 32, 8, 4, 2, 1) if args.height % p == 0)
+        pw = next(p for p in (24, 32, 16, 8, 4, 2, 1) if args.width % p == 0)
+        ae.y_patch_size = (ph, pw)
     ae.AE_only = bool(args.ae_only)
     ae.load_model = False
     ae.train_model = True
