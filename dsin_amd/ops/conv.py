@@ -10,9 +10,11 @@ geometry transforms:
     out[pixel][cout] = sum_k xbuf[mbase[pixel]+koff[k]] * w[cout][k];
   * backward-data is ANOTHER gather conv: stride-1 dilated conv of the
     zero-stuffed padded dy with spatially-rotated ci<->co-swapped weights;
-  * backward-weight is the conv_wrw kernel (fp32 atomics over pixel chunks);
-  * pad / stuff / flip / permute happen in torch, so their gradients
-    (cropping, un-stuffing) come from autograd for free.
+  * backward-weight is the conv_wrw kernel (per-pixel-chunk fp32 partial
+    slices reduced with one sum — no atomics);
+  * pad / zero-stuff and the rotated W panel are single fused kernels
+    (pad_stuff, wmat_make); the pad gradient (crop / un-stuff) is a
+    strided slice.
 
 bf16 compute with fp32 accumulate; fp32 master weights. Optional fused
 epilogue activation (relu / leaky-0.2) for the no-batchnorm convs (siNet).
@@ -76,7 +78,7 @@ def _wmat(w1: torch.Tensor, fp8: bool = False) -> torch.Tensor:
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
 
 
-def _wmat_rot(w1: torch.Tensor, co: int, ci: int, khw: int) -> torch.Tensor:
+def _wmat_rot(w1: torch.Tensor, khw: int) -> torch.Tensor:
     """Padded bf16 W panel for the backward-data gather conv: spatial taps
     reversed and cin/cout swapped — one kernel instead of the torch
     flip+permute+reshape+pad chain (~4 kernels per conv backward)."""
@@ -168,7 +170,7 @@ class _GatherConvFn(torch.autograd.Function):
                 dybuf = _padded_buf(dy, pe_w, pe_w, pe_h, pe_h, stride=stride)
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
                                  kh, kw, 1, dil, Hp, Wp)
-                dxbuf = ext_fwd(dybuf, _wmat_rot(w1, Co, Ci, kh * kw), None,
+                dxbuf = ext_fwd(dybuf, _wmat_rot(w1, kh * kw), None,
                                 mb2, ko2, Ci, Co * kh * kw, Hp, Wp, 0, 1)
 
         dw1 = None
@@ -386,7 +388,7 @@ class _GatherConv3dFn(torch.autograd.Function):
                                                kd, kh, kw)
                 assert (D2, H2, W2) == (Dp, Hp, Wp)
                 dxbuf = ext_fwd(dybuf.view(B, Co, -1, dybuf.shape[4]),
-                                _wmat_rot(w1, Co, Ci, kd * kh * kw), None,
+                                _wmat_rot(w1, kd * kh * kw), None,
                                 mb2, ko2, Ci, Co * kd * kh * kw, Dp * Hp, Wp,
                                 0, 1).view(B, Ci, Dp, Hp, Wp)
 

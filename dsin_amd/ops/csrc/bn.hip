@@ -79,20 +79,27 @@ __global__ void bn_stats_part_kernel(const bnbf16* __restrict__ y,
   }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ psum,   // (S,C)
-                                   const float* __restrict__ psum2,
-                                   float* __restrict__ mean,
-                                   float* __restrict__ rstd,
-                                   float* __restrict__ scale,
-                                   float* __restrict__ shift,
-                                   const float* __restrict__ gamma,
-                                   const float* __restrict__ beta,
-                                   float* __restrict__ rmean,
-                                   float* __restrict__ rvar,
-                                   int C, int S, float n, float momentum,
-                                   float eps, int training) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+// Applies the normalization AND derives scale/shift in-block from the
+// (S, C) partial sums (train) or the running stats (eval): the separate
+// bn_finalize launch (one tiny kernel per BN call, ~0.6 ms/step across the
+// model) is gone. The first block of each channel also publishes
+// mean/rstd for the backward and advances the EMA running stats.
+__global__ void bn_apply_kernel(const bnbf16* __restrict__ y,
+                                const bnbf16* __restrict__ res,  // or null
+                                bnbf16* __restrict__ out,
+                                const float* __restrict__ psum,  // (S,C)|null
+                                const float* __restrict__ psum2,
+                                const float* __restrict__ gamma,
+                                const float* __restrict__ beta,
+                                float* __restrict__ mean_out,
+                                float* __restrict__ rstd_out,
+                                float* __restrict__ rmean,
+                                float* __restrict__ rvar,
+                                int C, long long HW, int S, float n,
+                                float momentum, float eps, int training,
+                                int act) {
+  const int bc = blockIdx.y;
+  const int c = bc % C;
   float m, rs;
   if (training) {
     float t1 = 0.f, t2 = 0.f;
@@ -101,31 +108,24 @@ __global__ void bn_finalize_kernel(const float* __restrict__ psum,   // (S,C)
       t2 += psum2[sdx * C + c];
     }
     m = t1 / n;
-    float var = fmaxf(t2 / n - m * m, 0.f);
+    const float var = fmaxf(t2 / n - m * m, 0.f);
     rs = rsqrtf(var + eps);
-    const float unbias = (n > 1.f) ? var * n / (n - 1.f) : var;
-    rmean[c] = (1.f - momentum) * rmean[c] + momentum * m;
-    rvar[c] = (1.f - momentum) * rvar[c] + momentum * unbias;
+    if (blockIdx.x == 0 && bc < C && threadIdx.x == 0) {
+      mean_out[c] = m;
+      rstd_out[c] = rs;
+      const float unbias = (n > 1.f) ? var * n / (n - 1.f) : var;
+      rmean[c] = (1.f - momentum) * rmean[c] + momentum * m;
+      rvar[c] = (1.f - momentum) * rvar[c] + momentum * unbias;
+    }
   } else {
     m = rmean[c];
     rs = rsqrtf(rvar[c] + eps);
+    if (blockIdx.x == 0 && bc < C && threadIdx.x == 0) {
+      mean_out[c] = m;
+      rstd_out[c] = rs;
+    }
   }
-  mean[c] = m;
-  rstd[c] = rs;
-  const float sc = gamma[c] * rs;
-  scale[c] = sc;
-  shift[c] = beta[c] - m * sc;
-}
-
-__global__ void bn_apply_kernel(const bnbf16* __restrict__ y,
-                                const bnbf16* __restrict__ res,  // or null
-                                bnbf16* __restrict__ out,
-                                const float* __restrict__ scale,
-                                const float* __restrict__ shift,
-                                int C, long long HW, int act) {
-  const int bc = blockIdx.y;
-  const int c = bc % C;
-  const float sc = scale[c], sh = shift[c];
+  const float sc = gamma[c] * rs, sh = beta[c] - m * sc;
   const bnbf16* p = y + (long long)bc * HW;
   const bnbf16* q = res ? res + (long long)bc * HW : nullptr;
   bnbf16* o = out + (long long)bc * HW;
@@ -275,33 +275,17 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
   auto optsF = y.options().dtype(torch::kFloat32);
   auto mean = torch::empty({C}, optsF);
   auto rstd = torch::empty({C}, optsF);
-  auto scale = torch::empty({C}, optsF);
-  auto shift = torch::empty({C}, optsF);
   auto stream = at::cuda::getCurrentCUDAStream();
+  const int S = _spatial_chunks(HW);
+  float* psum = nullptr;
+  float* psum2 = nullptr;
+  torch::Tensor parts;
   if (training) {
-    const int S = _spatial_chunks(HW);
-    auto parts = torch::empty({2, S, C}, optsF);  // written fully: no fill
-    float* psum = parts.data_ptr<float>();
-    float* psum2 = psum + (long long)S * C;
+    parts = torch::empty({2, S, C}, optsF);  // written fully: no fill
+    psum = parts.data_ptr<float>();
+    psum2 = psum + (long long)S * C;
     hipLaunchKernelGGL(bn_stats_part_kernel, dim3(S, C), dim3(256), 0, stream,
                        (const bnbf16*)y.data_ptr(), psum, psum2, C, HW, B, S);
-    hipLaunchKernelGGL(bn_finalize_kernel, grid1d(C, 256), dim3(256), 0,
-                       stream, psum, psum2,
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       scale.data_ptr<float>(), shift.data_ptr<float>(),
-                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                       rmean.data_ptr<float>(), rvar.data_ptr<float>(), C, S,
-                       (float)((long long)B * HW), (float)momentum, (float)eps,
-                       1);
-  } else {
-    hipLaunchKernelGGL(bn_finalize_kernel, grid1d(C, 256), dim3(256), 0,
-                       stream, nullptr, nullptr, mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), scale.data_ptr<float>(),
-                       shift.data_ptr<float>(), gamma.data_ptr<float>(),
-                       beta.data_ptr<float>(), rmean.data_ptr<float>(),
-                       rvar.data_ptr<float>(), C, 0,
-                       (float)((long long)B * HW),
-                       (float)momentum, (float)eps, 0);
   }
   auto out = torch::empty_like(y);
   const bnbf16* resp = nullptr;
@@ -310,10 +294,14 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor y, torch::Tensor gamma,
     resp = (const bnbf16*)residual->data_ptr();
   }
   hipLaunchKernelGGL(bn_apply_kernel,
-                     dim3(_spatial_chunks(HW), B * C), dim3(256), 0, stream,
+                     dim3(S, B * C), dim3(256), 0, stream,
                      (const bnbf16*)y.data_ptr(), resp, (bnbf16*)out.data_ptr(),
-                     scale.data_ptr<float>(), shift.data_ptr<float>(), C, HW,
-                     (int)act);
+                     psum, psum2, gamma.data_ptr<float>(),
+                     beta.data_ptr<float>(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), rmean.data_ptr<float>(),
+                     rvar.data_ptr<float>(), C, HW, S,
+                     (float)((long long)B * HW), (float)momentum, (float)eps,
+                     training ? 1 : 0, (int)act);
   return {out, mean, rstd};
 }
 
