@@ -88,10 +88,14 @@ def main():
         model.decoder = model.decoder.to(memory_format=torch.channels_last)
         if model.sinet is not None:
             model.sinet = model.sinet.to(memory_format=torch.channels_last)
-    # hipGraph capture of in-graph RCCL collectives is unverified on this
-    # stack: capture only at world size 1, run eager when scaling out
+    # hipGraph capture includes the RCCL flat-grad all-reduce when
+    # distributed (verified single-rank on this stack; Trainer falls back
+    # to eager per-rank if capture throws, and a mixed captured/eager fleet
+    # still issues collectives in matching order). DSIN_NO_DIST_GRAPH=1
+    # restores the old capture-only-at-N=1 behavior.
     use_graph = (not args.no_graph and device.type == "cuda"
-                 and world_size() == 1)
+                 and (world_size() == 1
+                      or not os.environ.get("DSIN_NO_DIST_GRAPH")))
     trainer = Trainer(model, ae_config, pc_config, num_training_imgs=1576,
                      device=device, autocast_bf16=(args.dtype in ("bf16", "fp8")),
                      use_cuda_graph=use_graph,
