@@ -61,7 +61,8 @@ def _plan(device, Ci: int, Hp: int, Wp: int, kh: int, kw: int, stride: int,
     return p
 
 
-def _wmat(w1: torch.Tensor, fp8: bool = False) -> torch.Tensor:
+def _wmat(w1: torch.Tensor, fp8: bool = False,
+          direct: bool = False) -> torch.Tensor:
     """(Co, K) any-dtype -> bf16 (or e4m3-as-uint8) zero-padded to
     (Co, KP64+8); the zeros cancel the clamped out-of-range A gathers."""
     K = w1.shape[1]
@@ -73,16 +74,18 @@ def _wmat(w1: torch.Tensor, fp8: bool = False) -> torch.Tensor:
         out[:, :K] = w8
         return out
     if w1.is_cuda and hip_available():
-        return _require_ext("wmat_make")(w1.contiguous(), 1, False)
+        return _require_ext("wmat_make")(w1.contiguous(), 9 if direct else 1,
+                                         2 if direct else 0)
     KP = (K + 63) & ~63
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
 
 
-def _wmat_rot(w1: torch.Tensor, khw: int) -> torch.Tensor:
+def _wmat_rot(w1: torch.Tensor, khw: int, direct: bool = False) -> torch.Tensor:
     """Padded bf16 W panel for the backward-data gather conv: spatial taps
     reversed and cin/cout swapped — one kernel instead of the torch
-    flip+permute+reshape+pad chain (~4 kernels per conv backward)."""
-    return _require_ext("wmat_make")(w1.contiguous(), khw, True)
+    flip+permute+reshape+pad chain (~4 kernels per conv backward). direct:
+    mode-3 (chunk, tap, ci) layout for the 3x3 direct kernel."""
+    return _require_ext("wmat_make")(w1.contiguous(), khw, 3 if direct else 1)
 
 
 def _act_grad(dy: torch.Tensor, y_act, act: int) -> torch.Tensor:
@@ -141,14 +144,15 @@ class _GatherConvFn(torch.autograd.Function):
     * w1[co, (ci, r, s)]  (+bias, +activation). VALID over xbuf."""
 
     @staticmethod
-    def forward(ctx, xbuf, w1, bias, stride, dil, kh, kw, HO, WO, act):
+    def forward(ctx, xbuf, w1, bias, stride, dil, kh, kw, HO, WO, act,
+                direct):
         ext_fwd = _require_ext("conv_fwd")
         B, Ci, Hp, Wp = xbuf.shape
         Co, K = w1.shape
         mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
         bias32 = bias.float().contiguous() if bias is not None else None
-        y = ext_fwd(xbuf, _wmat(w1), bias32, mbase, koff, Co, K, HO, WO, act,
-                    stride)
+        y = ext_fwd(xbuf, _wmat(w1, direct=direct), bias32, mbase, koff, Co,
+                    K, HO, WO, act, stride, int(direct))
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act)
         return y
@@ -170,8 +174,10 @@ class _GatherConvFn(torch.autograd.Function):
                 dybuf = _padded_buf(dy, pe_w, pe_w, pe_h, pe_h, stride=stride)
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
                                  kh, kw, 1, dil, Hp, Wp)
-                dxbuf = ext_fwd(dybuf, _wmat_rot(w1, kh * kw), None,
-                                mb2, ko2, Ci, Co * kh * kw, Hp, Wp, 0, 1)
+                dirb = dil == 1 and kh == 3 and kw == 3 and Co % 64 == 0
+                dxbuf = ext_fwd(dybuf, _wmat_rot(w1, kh * kw, dirb), None,
+                                mb2, ko2, Ci, Co * kh * kw, Hp, Wp, 0, 1,
+                                int(dirb))
 
         dw1 = None
         if ctx.needs_input_grad[1]:
@@ -182,7 +188,8 @@ class _GatherConvFn(torch.autograd.Function):
 
         dbias = (dy.sum(dim=(0, 2, 3), dtype=torch.float32)
                  if has_bias else None)
-        return dxbuf, dw1, dbias, None, None, None, None, None, None, None
+        return (dxbuf, dw1, dbias, None, None, None, None, None, None, None,
+                None)
 
 
 class _GatherConvFP8Fn(torch.autograd.Function):
@@ -203,7 +210,7 @@ class _GatherConvFP8Fn(torch.autograd.Function):
         mbase, koff = _plan(x.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
         bias32 = bias.float().contiguous() if bias is not None else None
         y = ext_fwd(xbuf, _wmat(w1, fp8=True), bias32, mbase, koff, Co, K,
-                    HO, WO, act, stride)
+                    HO, WO, act, stride, 0)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act,
                     pads, stuff, x.shape, x.dtype)
@@ -239,7 +246,7 @@ class _GatherConvFP8Fn(torch.autograd.Function):
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
                                  kh, kw, 1, dil, Hp, Wp)
                 dxbuf = ext_fwd(dybuf, _wmat(wrot, fp8=True), None, mb2, ko2,
-                                Ci, Co * kh * kw, Hp, Wp, 0, 1)
+                                Ci, Co * kh * kw, Hp, Wp, 0, 1, 0)
                 _, _, H, W = xshape
                 dx = dxbuf[:, :, pt:pt + (H - 1) * stuff + 1:stuff,
                            pl:pl + (W - 1) * stuff + 1:stuff].to(xdtype)
@@ -278,8 +285,10 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
                                       stride, dilation, kh, kw, HO, WO, act,
                                       (padding, padding, padding, padding), 1)
     xbuf = _padded_buf(x, padding, padding, padding, padding)
+    direct = (stride == 1 and dilation == 1 and kh == 3 and kw == 3
+              and Ci % 64 == 0)
     return _GatherConvFn.apply(xbuf, w.reshape(Co, Ci * kh * kw), bias,
-                               stride, dilation, kh, kw, HO, WO, act)
+                               stride, dilation, kh, kw, HO, WO, act, direct)
 
 
 def conv_transpose2d(x: torch.Tensor, w: torch.Tensor,
@@ -305,7 +314,9 @@ def conv_transpose2d(x: torch.Tensor, w: torch.Tensor,
             (pl_w, pl_w + output_padding, pl_h, pl_h + output_padding), stride)
     xbuf = _padded_buf(x, pl_w, pl_w + output_padding,
                        pl_h, pl_h + output_padding, stride=stride)
-    return _GatherConvFn.apply(xbuf, w1, bias, 1, 1, kh, kw, HO, WO, act)
+    direct = kh == 3 and kw == 3 and Ci % 64 == 0
+    return _GatherConvFn.apply(xbuf, w1, bias, 1, 1, kh, kw, HO, WO, act,
+                               direct)
 
 
 def _act(y: torch.Tensor, act: int) -> torch.Tensor:
@@ -364,7 +375,7 @@ class _GatherConv3dFn(torch.autograd.Function):
         bias32 = bias.float().contiguous() if bias is not None else None
         # kernel sees a 2D problem: M = Do*Ho*Wo pixels, "WO" = Wo rows
         y = ext_fwd(xbuf.view(B, Ci, Dp * Hp, Wp), _wmat(w1), bias32, mbase,
-                    koff, Co, K, Do * Ho, Wo, act, 1)
+                    koff, Co, K, Do * Ho, Wo, act, 1, 0)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (kd, kh, kw, Do, Ho, Wo, bias is not None, act)
         return y.view(B, Co, Do, Ho, Wo)
@@ -390,7 +401,7 @@ class _GatherConv3dFn(torch.autograd.Function):
                 dxbuf = ext_fwd(dybuf.view(B, Co, -1, dybuf.shape[4]),
                                 _wmat_rot(w1, kd * kh * kw), None,
                                 mb2, ko2, Ci, Co * kd * kh * kw, Dp * Hp, Wp,
-                                0, 1).view(B, Ci, Dp, Hp, Wp)
+                                0, 1, 0).view(B, Ci, Dp, Hp, Wp)
 
         dw1 = None
         if ctx.needs_input_grad[1]:

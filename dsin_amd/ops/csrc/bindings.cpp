@@ -23,12 +23,12 @@ std::tuple<torch::Tensor, torch::Tensor> conv_tables(
 torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        c10::optional<torch::Tensor> bias, torch::Tensor mbase,
                        torch::Tensor koff, int64_t N, int64_t K, int64_t HO,
-                       int64_t WO, int64_t act, int64_t stride);
+                       int64_t WO, int64_t act, int64_t stride, int64_t direct);
 torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t WO, bool mcontig);
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act);
-torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, bool rot);
+torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, int64_t mode);
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor lr, torch::Tensor step,
                c10::optional<torch::Tensor> wd, double b1, double b2,

@@ -29,7 +29,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        c10::optional<torch::Tensor> bias,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t HO, int64_t WO, int64_t act,
-                       int64_t stride) {
+                       int64_t stride, int64_t direct) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(wmat);
   const bool fp8 = xbuf.scalar_type() == torch::kByte;  // raw e4m3 bytes
@@ -38,6 +38,33 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   TORCH_CHECK(wmat.scalar_type() == xbuf.scalar_type(), "x/w dtype mismatch");
   const int64_t B = xbuf.size(0);
   const int64_t M = HO * WO;
+  if (direct && !fp8) {
+    // stride-1 3x3, Ci%64==0: direct LDS-halo kernel (9x less gather
+    // traffic than the im2col path); caller built wmat in mode-2/3 layout
+    const int64_t Ci = xbuf.size(1);
+    const int64_t KPd = (K + 63) & ~63;
+    TORCH_CHECK(K == Ci * 9 && Ci % 64 == 0 && stride == 1,
+                "direct conv gate mismatch");
+    TORCH_CHECK(wmat.size(1) == KPd + CONV_AP, "direct wmat stride mismatch");
+    auto out = torch::empty({B, N, HO, WO},
+                            xbuf.options().dtype(torch::kBFloat16));
+    const float* bp = nullptr;
+    if (bias.has_value()) {
+      CHECK_CUDA_CONTIG(bias.value());
+      bp = bias->data_ptr<float>();
+    }
+    const int Hp = (int)xbuf.size(2), Wp = (int)xbuf.size(3);
+    dim3 grid(((HO + 7) / 8) * ((WO + 7) / 8), (N + 63) / 64, B);
+    size_t lds = (size_t)2 * 10 * 11 * 64 * 2;
+    hipLaunchKernelGGL(conv3x3_direct_kernel, grid, dim3(256), lds,
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
+                       (const cvbf16*)xbuf.data_ptr(),
+                       (const cvbf16*)wmat.data_ptr(), bp,
+                       (cvbf16*)out.data_ptr(), (int)Ci, Hp, Wp, (int)N,
+                       (int)HO, (int)WO, (int)KPd, xbuf.stride(0),
+                       (long long)N * M, (int)act);
+    return out;
+  }
   const int64_t KP = (K + 63) & ~63;  // 64-chunk padded; wmat zero-padded
   TORCH_CHECK(wmat.size(1) == KP + CONV_AP, "wmat row stride mismatch");
   if (WO < 8) stride = 0;  // scalar staging path for very narrow outputs
@@ -159,9 +186,10 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act) {
   return out;
 }
 
-torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, bool rot) {
+torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, int64_t mode) {
   CHECK_CUDA_CONTIG(w1);
   const int64_t N = w1.size(0), K = w1.size(1);
+  const bool rot = (mode & 1) != 0;
   const int64_t rows = rot ? K / khw : N;
   const int64_t kout = rot ? N * khw : K;
   const int64_t KPA = ((kout + 63) & ~63) + CONV_AP;
@@ -174,13 +202,13 @@ torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, bool rot) {
     hipLaunchKernelGGL((wmat_make_kernel<float>), dim3(grid), dim3(256), 0,
                        stream, w1.data_ptr<float>(), (cvbf16*)out.data_ptr(),
                        (int)rows, (int)kout, (int)K, (int)khw, (int)KPA,
-                       rot ? 1 : 0);
+                       (int)mode);
   } else {
     TORCH_CHECK(w1.scalar_type() == torch::kBFloat16, "wmat_make: fp32/bf16");
     hipLaunchKernelGGL((wmat_make_kernel<cvbf16>), dim3(grid), dim3(256), 0,
                        stream, (const cvbf16*)w1.data_ptr(),
                        (cvbf16*)out.data_ptr(), (int)rows, (int)kout, (int)K,
-                       (int)khw, (int)KPA, rot ? 1 : 0);
+                       (int)khw, (int)KPA, (int)mode);
   }
   return out;
 }

@@ -309,6 +309,159 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   }
 }
 
+// Direct 3x3 stride-1 convolution (the resblock / backward-data hot
+// shapes: Ci a multiple of 64, dilation 1). The gather-GEMM kernel above
+// re-reads every input element ~9 times through L2 (once per filter tap);
+// this kernel stages each 8x8-output tile's 10x10 input halo ONCE per
+// 64-channel chunk in LDS (pixel-major, channels contiguous) and runs all
+// nine taps out of LDS: 9x less gather traffic and 72 MFMAs between
+// barriers (one barrier per ci-chunk). W panel layout is (chunk, tap, ci)
+// -- wmat_make mode 2/3.
+//   out[n][oy0+py][ox0+px] = sum_{ci,r,s} xpad[ci][oy0+py+r][ox0+px+s]
+//                                        * w[n][(ci,r,s)]
+__global__ __launch_bounds__(256)
+void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
+                           const cvbf16* __restrict__ wmat, // (N, KP+AP)
+                           const float* __restrict__ bias,
+                           cvbf16* __restrict__ out,        // (N, HO, WO)
+                           int Ci, int Hp, int Wp, int N, int HO, int WO,
+                           int KP, long long x_img_stride,
+                           long long o_img_stride, int act) {
+  constexpr int CIC = 64;           // input channels per LDS chunk
+  constexpr int XT = 11;            // LDS tile row stride in pixels (odd:
+                                    // breaks 2-row bank aliasing)
+  constexpr int TB = 10 * XT * CIC; // bf16 elems per LDS buffer
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  cvbf16* As = reinterpret_cast<cvbf16*>(smem);  // 2 x [10][XT][CIC]
+  char* As8 = reinterpret_cast<char*>(As);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int tx = (WO + 7) >> 3;
+  const int oy0 = (blockIdx.x / tx) * 8;
+  const int ox0 = (blockIdx.x % tx) * 8;
+  const long long img = blockIdx.z;
+  const cvbf16* x = xpad + img * x_img_stride;
+
+  const int colL = lane & 15;
+  const int kgrp = lane >> 4;
+  const int n0 = blockIdx.y * CONV_TN + wid * 16;
+  const int ncol = n0 + colL;
+  const int WSTRIDE = KP + CONV_AP;
+  const cvbf16* wrow = wmat + (long long)(ncol < N ? ncol : 0) * WSTRIDE;
+
+  cv_f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                     {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+
+  // LDS address of (pixel p = y*XT+x, channel ci), with a 16B-granule XOR
+  // swizzle on the channel group so column reads spread across banks
+  auto aoff = [&](int p, int ci) -> int {
+    return p * (CIC * 2) + ((((ci >> 3) ^ (p & 7)) << 4) | ((ci & 7) << 1));
+  };
+
+  typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+  // stage chunk c's 10x10xCIC halo tile into LDS buffer buf
+  auto stage = [&](int c, int buf) {
+    char* dst = As8 + buf * TB * 2;
+    for (int ridx = tid; ridx < CIC * 10; ridx += 256) {
+      const int ci = ridx / 10, y = ridx % 10;
+      const int yy = min(oy0 + y, Hp - 1);  // bottom edge tiles: clamped
+      const cvbf16* g = x + ((long long)(c * CIC + ci) * Hp + yy) * Wp + ox0;
+      const u16x8 a = *reinterpret_cast<const u16x8*>(g);
+      const u16x8 b = *reinterpret_cast<const u16x8*>(g + 8);
+#pragma unroll
+      for (int xi = 0; xi < 10; ++xi) {
+        unsigned short v = xi < 8 ? a[xi] : b[xi - 8];
+        *reinterpret_cast<cvbf16*>(&dst[aoff(y * XT + xi, ci)]) =
+            *reinterpret_cast<cvbf16*>(&v);
+      }
+    }
+  };
+
+  // my 16 output pixels per MFMA row-subtile: row l = mi*16 + colL
+  // -> pixel (py, px) = ((mi*16+colL) >> 3, (mi*16+colL) & 7)
+  const int nchunks = Ci / CIC;
+  // W prefetch: one 64-k tap group (2 MFMA k-steps) ahead — small register
+  // sets (2 x 2 x b128) keep total pressure low enough for 3+ waves/SIMD
+  cv_bf16x8 wA[2], wB[2];
+  auto load_w = [&](int kbase, cv_bf16x8* wset) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+      wset[j] = *reinterpret_cast<const cv_bf16x8*>(
+          &wrow[kbase + j * 32 + kgrp * 8]);
+  };
+
+  stage(0, 0);
+  load_w(0, wA);
+  if (nchunks > 1) stage(1, 1);
+  __syncthreads();
+
+  // chunk body with COMPILE-TIME chunk parity: the W register-set rotation
+  // (tap group index is 9c+tap, parity (c+tap)&1) must fold to direct
+  // register references — a runtime-selected pointer would spill both sets
+  auto run_chunk = [&](auto codd, int c) {
+    constexpr int CO = decltype(codd)::value;
+    const char* cur = As8 + CO * TB * 2;
+    const int kb = c * 9 * CIC;
+#pragma unroll
+    for (int tap = 0; tap < 9; ++tap) {
+      cv_bf16x8* wcur = ((CO + tap) & 1) ? wB : wA;
+      cv_bf16x8* wnxt = ((CO + tap) & 1) ? wA : wB;
+      if (tap < 8) load_w(kb + (tap + 1) * CIC, wnxt);
+      else if (c + 1 < nchunks) load_w(kb + 9 * CIC, wnxt);
+      const int r = tap / 3, sx = tap % 3;
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int cio = (kk << 5) + kgrp * 8;
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          const int m = mi * 16 + colL;
+          const int p = ((m >> 3) + r) * XT + (m & 7) + sx;
+          const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
+              &cur[aoff(p, cio)]);
+          acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, wcur[kk], acc[mi], 0, 0, 0);
+        }
+      }
+    }
+  };
+  constexpr std::integral_constant<int, 0> C0{};
+  constexpr std::integral_constant<int, 1> C1{};
+  for (int c = 0; c < nchunks; ++c) {
+    if (c & 1) run_chunk(C1, c);
+    else run_chunk(C0, c);
+    if (c + 1 < nchunks) {
+      if (c + 2 < nchunks) {
+        __syncthreads();        // buffer (c&1) free only after all waves
+        stage(c + 2, c & 1);    // finish the MFMA phase above
+      }
+      __syncthreads();
+    }
+  }
+
+  // epilogue: D rows are tile pixels; row = mi*16 + kgrp*4 + reg
+  const float bv = (bias != nullptr && ncol < N) ? bias[ncol] : 0.f;
+  cvbf16* o = out + img * o_img_stride +
+              (long long)(ncol < N ? ncol : 0) * HO * WO;
+  if (ncol < N) {
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = mi * 16 + kgrp * 4 + reg;
+        const int oy = oy0 + (m >> 3), ox = ox0 + (m & 7);
+        if (oy < HO && ox < WO) {
+          float v = acc[mi][reg] + bv;
+          if (act == 1) v = fmaxf(v, 0.f);
+          else if (act == 2) v = fmaxf(v, 0.2f * v);
+          o[(long long)oy * WO + ox] = cvf2b(v);
+        }
+      }
+    }
+  }
+}
+
 // dW[co][k] += sum over the workgroup's pixel chunk of dy[co][m]*A[m][k].
 // GEMM roles: A' (M'=filter taps K) gathered rows, B' = dy columns.
 // Tile: M'64 (taps) x N'64 (couts), K' = pixels chunked by 32.
@@ -452,24 +605,32 @@ __global__ void act_bwd_kernel(const T* __restrict__ dy,
 
 // builds the zero-padded bf16 W panel for conv_fwd in ONE kernel, replacing
 // the torch pad+contiguous (forward) or flip+permute+reshape+pad chain
-// (backward-data "rotated" weights, ~4 kernels per conv backward).
-//   rot=0: out[n][k] = w1[n][k]                       (N=Co rows, K taps)
-//   rot=1: out[ci][co*khw + t] = w1[co][ci*khw + khw-1-t]
-// (full multi-radix tap reversal == flip of every spatial dim), k >= K -> 0.
+// (backward-data "rotated" weights, ~4 kernels per conv backward). mode:
+//   0: out[n][k] = w1[n][k]                       (N=Co rows, K taps)
+//   1: out[ci][co*khw + t] = w1[co][ci*khw + khw-1-t]
+//      (full multi-radix tap reversal == flip of every spatial dim)
+//   2: direct-conv layout k' = (ci/64)*64*khw + t*64 + ci%64 (channels
+//      contiguous per tap per 64-chunk — see conv3x3_direct_kernel)
+//   3: mode 1 + mode 2 (rotated weights in direct layout)
+// k >= K -> 0.
 template <typename T>
 __global__ void wmat_make_kernel(const T* __restrict__ w1,
                                  cvbf16* __restrict__ out,
                                  int rows, int kout, int kin, int khw,
-                                 int KPA, int rot) {
+                                 int KPA, int mode) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   long long total = (long long)rows * KPA;
   long long gstride = (long long)gridDim.x * blockDim.x;
   for (; i < total; i += gstride) {
     const int n = (int)(i / KPA);
-    const int k = (int)(i % KPA);
+    int k = (int)(i % KPA);
     float v = 0.f;
     if (k < kout) {
-      if (rot) {
+      if (mode >= 2) {   // decode direct layout: k' -> (c, tap, ci_in_chunk)
+        const int chunk = k / (64 * khw), rm = k % (64 * khw);
+        k = (chunk * 64 + rm % 64) * khw + rm / 64;
+      }
+      if (mode & 1) {
         const int co = k / khw, t = k % khw;
         v = (float)w1[(long long)co * kin + n * khw + (khw - 1 - t)];
       } else {

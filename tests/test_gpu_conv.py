@@ -244,3 +244,45 @@ def test_conv_fp8_path(dev):
         assert float(relt) < 0.15, float(relt)
     finally:
         dconv.set_compute_dtype("bf16")
+
+
+@pytest.mark.gpu
+def test_conv3x3_direct_path(dev):
+    """Direct LDS-halo 3x3 kernel (stride 1, Ci%64==0) vs torch fp32:
+    forward + input/weight grads, odd sizes (edge-tile masking), and the
+    conv-transpose route (stuffed stride-1 3x3 with Co%64==0 backward)."""
+    from dsin_amd.ops import conv as dconv
+    torch.manual_seed(1)
+    for Ci, Co, H, W in [(64, 48, 24, 40), (128, 128, 19, 29)]:
+        x = torch.randn(2, Ci, H, W, device=dev)
+        w = torch.randn(Co, Ci, 3, 3, device=dev) / (3 * Ci ** 0.5)
+        x1 = x.clone().requires_grad_(True)
+        w1 = w.clone().requires_grad_(True)
+        y = dconv.conv2d(x1, w1, None, 1, 1, 1)
+        x2 = x.clone().requires_grad_(True)
+        w2 = w.clone().requires_grad_(True)
+        yr = F.conv2d(x2, w2, None, stride=1, padding=1)
+        assert torch.allclose(y.float(), yr, atol=0.05, rtol=0.05), \
+            (Ci, Co, (y.float() - yr).abs().max().item())
+        g = torch.randn_like(yr)
+        y.backward(g.to(y.dtype))
+        yr.backward(g)
+        assert torch.allclose(x1.grad.float(), x2.grad, atol=0.08, rtol=0.08)
+        dwmax = w2.grad.abs().max().item()
+        assert torch.allclose(w1.grad.float(), w2.grad,
+                              atol=0.02 * max(dwmax, 1.0), rtol=0.05)
+    # conv-transpose: backward-data of the stuffed conv takes the direct
+    # path when Co%64==0
+    xt = torch.randn(1, 128, 10, 14, device=dev).requires_grad_(True)
+    wt = torch.randn(128, 64, 3, 3, device=dev) * 0.05
+    wt1 = wt.clone().requires_grad_(True)
+    yt = dconv.conv_transpose2d(xt, wt1, None, 2, 1, 1)
+    xr = xt.detach().clone().requires_grad_(True)
+    wr = wt.clone().requires_grad_(True)
+    ytr = F.conv_transpose2d(xr, wr, None, stride=2, padding=1,
+                             output_padding=1)
+    assert torch.allclose(yt.float(), ytr, atol=0.05, rtol=0.05)
+    gt = torch.randn_like(ytr)
+    yt.backward(gt.to(yt.dtype))
+    ytr.backward(gt)
+    assert torch.allclose(xt.grad.float(), xr.grad, atol=0.08, rtol=0.08)
