@@ -18,7 +18,7 @@ from __future__ import annotations
 import os
 import queue
 import threading
-from typing import Iterator, List, Optional, Tuple
+from typing import Iterator, List, Tuple
 
 import numpy as np
 import torch

@@ -13,7 +13,6 @@ Mirror of /root/reference/src/Distortions_imgcomp.py. Semantics kept exactly:
 
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch

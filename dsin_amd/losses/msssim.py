@@ -11,7 +11,6 @@ prod(cs[:4]**w[:4]) * ssim[4]**w[4] (:164-186). Inputs NCHW in 0..255.
 
 from __future__ import annotations
 
-import math
 from typing import Tuple
 
 import torch

@@ -21,7 +21,6 @@ correlates *decoded* x against *decoded* y.
 
 from __future__ import annotations
 
-from contextlib import nullcontext
 from typing import Dict, NamedTuple, Optional
 
 import torch

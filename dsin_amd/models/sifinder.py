@@ -15,7 +15,6 @@ autograd entirely.
 
 from __future__ import annotations
 
-from typing import Tuple
 
 import torch
 

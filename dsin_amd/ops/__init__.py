@@ -12,13 +12,11 @@ CPU tensors always use the reference implementations in
 
 from __future__ import annotations
 
-import os
 from typing import Optional, Tuple
 
 import torch
 
 from . import reference as ref
-from ..constants import LOG2_E
 
 _EXT = None
 _EXT_ERR: Optional[str] = None

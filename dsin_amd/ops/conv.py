@@ -23,7 +23,7 @@ CPU path falls back to torch.nn.functional (the numerics oracle).
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn.functional as F

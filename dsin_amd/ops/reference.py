@@ -9,7 +9,7 @@ documents the reference-repo semantics it mirrors (file:line cites are into
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn.functional as F

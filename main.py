@@ -18,7 +18,7 @@ import torch
 from dsin_amd import config as config_mod
 from dsin_amd.data import make_dataset
 from dsin_amd.models import DSIN
-from dsin_amd.parallel import init_distributed, is_distributed, rank
+from dsin_amd.parallel import init_distributed, rank
 from dsin_amd.training import Trainer, checkpoint
 from dsin_amd.utils import MetricsLogger, loss_list_saver, save_test_img
 
