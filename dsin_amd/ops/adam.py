@@ -44,7 +44,10 @@ class FusedAdam:
             n = p.numel()
             self.flat_p[ofs:ofs + n].copy_(p.data.reshape(-1))
             p.data = self.flat_p[ofs:ofs + n].view(p.shape)
-            p.grad = self.flat_g[ofs:ofs + n].view(p.shape)
+            # grads are NOT repointed into flat_g: with p.grad left None,
+            # autograd STEALS each produced gradient (zero accumulate-add
+            # kernels, ~330 of them per DSIN step) and gather_grads() moves
+            # them into the flat buffer with one _foreach_copy_ sweep.
             self._slices.append((ofs, n))
             ofs += n
 
@@ -57,9 +60,26 @@ class FusedAdam:
         self.param_groups = [{"lr": self.lr_t, "params": self.params}]
         self.defaults = {"lr": float(lr)}
 
-    def zero_grad(self, set_to_none: bool = False):
-        # set_to_none would break the flat-buffer invariant; always zero
-        self.flat_g.zero_()
+    def zero_grad(self, set_to_none: bool = True):
+        # gather_grads() overwrites flat_g, so only the python-side grad
+        # handles need dropping (autograd then steals fresh tensors)
+        for p in self.params:
+            p.grad = None
+
+    @torch.no_grad()
+    def gather_grads(self):
+        """Copy this step's stolen .grad tensors into the flat comm/step
+        buffer (one horizontally-fused kernel); missing grads zero their
+        slice. Must run after backward, before any all-reduce / step()."""
+        dsts, srcs = [], []
+        for p, (ofs, n) in zip(self.params, self._slices):
+            if p.grad is None:
+                self.flat_g[ofs:ofs + n].zero_()
+            else:
+                dsts.append(self.flat_g[ofs:ofs + n])
+                srcs.append(p.grad.reshape(-1))
+        if dsts:
+            torch._foreach_copy_(dsts, srcs)
 
     def set_weight_decay(self, param: torch.nn.Parameter, factor: float) -> bool:
         """Register a classic L2 term factor/2*sum(p^2) for `param`: its

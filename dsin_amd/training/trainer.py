@@ -156,19 +156,25 @@ class Trainer:
     def _step_inner(self, x: torch.Tensor, y: Optional[torch.Tensor]):
         """zero-grad + forward + backward + both optimizer steps. This is
         what gets graph-captured; it must stay free of host syncs."""
-        self.opt_ae.zero_grad(set_to_none=not self.use_cuda_graph)
-        self.opt_pc.zero_grad(set_to_none=not self.use_cuda_graph)
+        self.opt_ae.zero_grad(set_to_none=not self.use_cuda_graph
+                              or self._fused)
+        self.opt_pc.zero_grad(set_to_none=not self.use_cuda_graph
+                              or self._fused)
         self.reducer.prepare()
         with self._autocast():
             out = self.model.train_losses(x, y)
         out["loss"].backward()
+        if self._fused:
+            # stolen per-tensor grads -> flat buffers (one fused copy)
+            self.opt_ae.gather_grads()
+            self.opt_pc.gather_grads()
         if self._fused and is_distributed():
             import torch.distributed as dist
             ws = float(dist.get_world_size())
             for opt in (self.opt_ae, self.opt_pc):
                 dist.all_reduce(opt.flat_g)
                 opt.flat_g.div_(ws)
-        else:
+        elif not self._fused:
             self.reducer.finalize()
         self.opt_ae.step()
         self.opt_pc.step()
