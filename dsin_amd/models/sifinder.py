@@ -26,10 +26,13 @@ class SiFinder(torch.nn.Module):
         super().__init__()
         self.ph, self.pw = (int(v) for v in config.y_patch_size)
         self.use_mask = bool(config.use_gauss_mask)
-        if config.use_L2andLAB:
-            raise NotImplementedError(
-                "use_L2andLAB (L2+LAB search mode) is not implemented in the "
-                "MI355X rebuild; the shipped reference configs use Pearson+H1H2H3.")
+        self.l2lab = bool(config.use_L2andLAB)
+        if self.l2lab:
+            import warnings
+            warnings.warn(
+                "use_L2andLAB runs on the torch path (correct but slower "
+                "than the streaming HIP NCC kernel used for the shipped "
+                "Pearson+H1H2H3 mode)")
 
     @torch.no_grad()
     def forward(self, x_dec: torch.Tensor, y_orig: torch.Tensor,
@@ -38,6 +41,7 @@ class SiFinder(torch.nn.Module):
         outs = []
         for n in range(x_dec.shape[0]):
             y_syn, _, _ = ops.ncc_search(x_dec[n], y_dec[n], y_orig[n],
-                                         self.ph, self.pw, self.use_mask)
+                                         self.ph, self.pw, self.use_mask,
+                                         l2lab=self.l2lab)
             outs.append(y_syn)
         return torch.stack(outs).to(x_dec.dtype)

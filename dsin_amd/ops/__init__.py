@@ -143,16 +143,20 @@ def bitcost_ce(logits: torch.Tensor, symbols: torch.Tensor) -> torch.Tensor:
 
 @torch.no_grad()
 def ncc_search(x_dec: torch.Tensor, y_dec: torch.Tensor, y_orig: torch.Tensor,
-               ph: int, pw: int, use_mask: bool = True
+               ph: int, pw: int, use_mask: bool = True, l2lab: bool = False
                ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-    """Single-image SI search: (3,Hx,Wx)x3 -> (y_syn (3,Hx,Wx), rows, cols)."""
-    if x_dec.is_cuda:
+    """Single-image SI search: (3,Hx,Wx)x3 -> (y_syn (3,Hx,Wx), rows, cols).
+    l2lab selects the reference's use_L2andLAB mode (LAB transform +
+    argmin of squared L2) — served by the torch path on every device; the
+    streaming HIP kernel covers the Pearson+H1H2H3 mode of the shipped
+    configs."""
+    if x_dec.is_cuda and not l2lab:
         fn = _require_ext("ncc_search")
         y_syn, rows, cols = fn(x_dec.contiguous().float(), y_dec.contiguous().float(),
                                y_orig.contiguous().float(), ph, pw, use_mask)
         return y_syn, rows, cols
     return ref.ncc_search_ref(x_dec.float(), y_dec.float(), y_orig.float(),
-                              ph, pw, use_mask)
+                              ph, pw, use_mask, l2lab=l2lab)
 
 
 # re-exports used across the package

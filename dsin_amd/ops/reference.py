@@ -204,9 +204,37 @@ def gaussian_mask_value(num_patches_w: int, ph: int, pw: int, H: int, W: int,
     return g.to(device=device, dtype=dtype)
 
 
+def rgb_to_lab_ref(x: torch.Tensor) -> torch.Tensor:
+    """CIELAB transform used by the reference's use_L2andLAB search mode
+    (src/siFinder.py:157-195; applied to the tensors AS-IS, i.e. in the
+    raw pixel scale the caller passes — the reference does not rescale
+    before the sRGB gamma expansion). x: (..., 3, H, W)."""
+    px = x.movedim(-3, -1).reshape(-1, 3).float()
+    lin = torch.where(px <= 0.04045, px / 12.92,
+                      (((px + 0.055) / 1.055).clamp(min=0.0)) ** 2.4)
+    rgb_to_xyz = _cached_const(
+        ((0.412453, 0.212671, 0.019334),
+         (0.357580, 0.715160, 0.119193),
+         (0.180423, 0.072169, 0.950227)), x.device, torch.float32, (3, 3))
+    xyz = lin @ rgb_to_xyz
+    xyz = xyz * _cached_const((1.0 / 0.950456, 1.0, 1.0 / 1.088754),
+                              x.device, torch.float32, (1, 3))
+    eps3 = (6.0 / 29.0) ** 3
+    f = torch.where(xyz <= eps3,
+                    xyz / (3 * (6.0 / 29.0) ** 2) + 4.0 / 29.0,
+                    xyz.clamp(min=0.0) ** (1.0 / 3.0))
+    f_to_lab = _cached_const(
+        ((0.0, 500.0, 0.0), (116.0, -500.0, 200.0), (0.0, 0.0, -200.0)),
+        x.device, torch.float32, (3, 3))
+    lab = f @ f_to_lab + _cached_const((-16.0, 0.0, 0.0), x.device,
+                                       torch.float32, (1, 3))
+    return lab.reshape(*x.movedim(-3, -1).shape).movedim(-1, -3)
+
+
 def ncc_search_ref(x_dec: torch.Tensor, y_dec: torch.Tensor, y_orig: torch.Tensor,
                    ph: int, pw: int, use_mask: bool = True,
-                   eps: float = 1e-10) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+                   eps: float = 1e-10,
+                   l2lab: bool = False) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Full SI search for ONE image (the reference loops images at batch 1,
     src/siFull_img.py:15-39). All inputs (3, H, W) [y_* may be larger than
     x_dec in general; equal here]. Returns (y_syn (3,Hx,Wx), rows (P,), cols (P,)).
@@ -222,26 +250,35 @@ def ncc_search_ref(x_dec: torch.Tensor, y_dec: torch.Tensor, y_orig: torch.Tenso
     c, hx, wx = x_dec.shape
     _, hy, wy = y_dec.shape
     patches = extract_patches(x_dec, ph, pw)                    # (P,3,ph,pw)
-    q = _h1h2h3(_sifinder_norm(patches))                        # (P,3,ph,pw)
-    r = _h1h2h3(_sifinder_norm(y_dec)).unsqueeze(0)             # (1,3,Hy,Wy)
+    if l2lab:
+        # L2+LAB mode (src/siFinder.py:13-17): LAB transform of the RAW
+        # tensors, no fixed-stats normalization
+        q = rgb_to_lab_ref(patches)
+        r = rgb_to_lab_ref(y_dec).unsqueeze(0)
+    else:
+        q = _h1h2h3(_sifinder_norm(patches))                    # (P,3,ph,pw)
+        r = _h1h2h3(_sifinder_norm(y_dec)).unsqueeze(0)         # (1,3,Hy,Wy)
 
     n = float(ph * pw * 3)
     xy = F.conv2d(r, q)[0]                                      # (P,Hc,Wc)
     ones = r.new_ones(1, 3, ph, pw)
     sum_y = F.conv2d(r, ones)[0, 0]                             # (Hc,Wc)
     sum_y2 = F.conv2d(r * r, ones)[0, 0]
-    y_mean = sum_y / n
     sum_x = q.sum(dim=(1, 2, 3))                                # (P,)
     sum_x2 = (q * q).sum(dim=(1, 2, 3))
-    x_mean = sum_x / n
-
-    num = xy - y_mean.unsqueeze(0) * sum_x.view(-1, 1, 1) \
-        - sum_y.unsqueeze(0) * x_mean.view(-1, 1, 1) \
-        + n * (x_mean.view(-1, 1, 1) * y_mean.unsqueeze(0))
-    den_x = sum_x2 - 2 * x_mean * sum_x + n * x_mean ** 2       # (P,)
-    den_y = sum_y2 - 2 * y_mean * sum_y + n * y_mean ** 2       # (Hc,Wc)
-    den = den_y.unsqueeze(0) * den_x.view(-1, 1, 1)
-    ncc = num / torch.sqrt(den + eps)
+    if l2lab:
+        # squared L2 distance per window (src/siFinder.py:102-103)
+        ncc = (sum_x2.view(-1, 1, 1) - 2 * xy + sum_y2.unsqueeze(0))
+    else:
+        y_mean = sum_y / n
+        x_mean = sum_x / n
+        num = xy - y_mean.unsqueeze(0) * sum_x.view(-1, 1, 1) \
+            - sum_y.unsqueeze(0) * x_mean.view(-1, 1, 1) \
+            + n * (x_mean.view(-1, 1, 1) * y_mean.unsqueeze(0))
+        den_x = sum_x2 - 2 * x_mean * sum_x + n * x_mean ** 2   # (P,)
+        den_y = sum_y2 - 2 * y_mean * sum_y + n * y_mean ** 2   # (Hc,Wc)
+        den = den_y.unsqueeze(0) * den_x.view(-1, 1, 1)
+        ncc = num / torch.sqrt(den + eps)
 
     if use_mask:
         mask = gaussian_mask_value(wx // pw, ph, pw, hx, wx, x_dec.device, x_dec.dtype)
@@ -250,7 +287,9 @@ def ncc_search_ref(x_dec: torch.Tensor, y_dec: torch.Tensor, y_orig: torch.Tenso
         ncc = ncc * mask
 
     p_count, hc, wc = ncc.shape
-    flat_idx = ncc.view(p_count, -1).argmax(dim=1)
+    # L2 mode picks the MINIMUM distance (src/siFinder.py:28-31)
+    flat = ncc.view(p_count, -1)
+    flat_idx = flat.argmin(dim=1) if l2lab else flat.argmax(dim=1)
     rows = torch.div(flat_idx, wc, rounding_mode="floor")
     cols = flat_idx % wc
 

@@ -1,6 +1,7 @@
 import torch
 import pytest
 
+from dsin_amd import ops
 from dsin_amd.ops.reference import (assemble_patches, extract_patches,
                                     gaussian_mask_value, ncc_search_ref)
 
@@ -71,3 +72,29 @@ def test_sifinder_module_batch(small_ae_config):
     y = torch.rand(2, 3, 64, 96) * 255
     out = sf(x, y, y)
     assert out.shape == x.shape
+
+
+def test_l2lab_mode_planted_patch():
+    """use_L2andLAB search (reference src/siFinder.py:13-31, 102-103,
+    145-195): LAB transform, squared-L2 score, argmin. A patch of y copied
+    into x must match itself at its own location (no mask)."""
+    torch.manual_seed(5)
+    H, W, ph, pw = 40, 48, 8, 8
+    y = torch.rand(3, H, W) * 255
+    x = torch.rand(3, H, W) * 255
+    # plant y's (16, 24) patch at x patch-slot (1, 2)
+    x[:, 8:16, 16:24] = y[:, 16:24, 24:32]
+    y_syn, rows, cols = ops.ncc_search(x, y, y, ph, pw, use_mask=False,
+                                       l2lab=True)
+    p = (8 // ph) * (W // pw) + 16 // pw
+    assert int(rows[p]) == 16 and int(cols[p]) == 24
+    assert torch.allclose(y_syn[:, 8:16, 16:24], y[:, 16:24, 24:32])
+
+
+def test_l2lab_differs_from_pearson():
+    torch.manual_seed(6)
+    x = torch.rand(3, 32, 32) * 255
+    y = torch.rand(3, 32, 32) * 255
+    a, _, _ = ops.ncc_search(x, y, y, 8, 8, use_mask=True, l2lab=False)
+    b, _, _ = ops.ncc_search(x, y, y, 8, 8, use_mask=True, l2lab=True)
+    assert a.shape == b.shape
