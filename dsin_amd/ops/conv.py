@@ -152,7 +152,7 @@ class _GatherConvFn(torch.autograd.Function):
         mbase, koff = _plan(xbuf.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
         bias32 = bias.float().contiguous() if bias is not None else None
         y = ext_fwd(xbuf, _wmat(w1, direct=direct), bias32, mbase, koff, Co,
-                    K, HO, WO, act, stride, int(direct))
+                    K, HO, WO, act, stride, int(direct), 0)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act)
         return y
@@ -171,13 +171,25 @@ class _GatherConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             with torch.no_grad():
                 pe_h, pe_w = (kh - 1) * dil, (kw - 1) * dil
-                dybuf = _padded_buf(dy, pe_w, pe_w, pe_h, pe_h, stride=stride)
-                mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
+                dir3 = dil == 1 and kh == 3 and kw == 3 and Co % 64 == 0
+                vp_ok = dir3 and stride == 1  # no zero-stuffing to emulate
+                dybuf = (dy if vp_ok else
+                         _padded_buf(dy, pe_w, pe_w, pe_h, pe_h,
+                                     stride=stride))
+                mb2, ko2 = _plan(dy.device, Co,
+                                 dy.shape[2] * stride + 2 * pe_h - stride + 1,
+                                 dy.shape[3] * stride + 2 * pe_w - stride + 1,
                                  kh, kw, 1, dil, Hp, Wp)
-                dirb = dil == 1 and kh == 3 and kw == 3 and Co % 64 == 0
-                dxbuf = ext_fwd(dybuf, _wmat_rot(w1, kh * kw, dirb), None,
-                                mb2, ko2, Ci, Co * kh * kw, Hp, Wp, 0, 1,
-                                int(dirb))
+                if vp_ok:
+                    # virtual pad: the direct kernel zero-masks the halo,
+                    # so dy needs no padded buffer at all
+                    dxbuf = ext_fwd(dy, _wmat_rot(w1, kh * kw, True), None,
+                                    mb2, ko2, Ci, Co * kh * kw, Hp, Wp, 0,
+                                    1, 1, kh - 1)
+                else:
+                    dxbuf = ext_fwd(dybuf, _wmat_rot(w1, kh * kw, dir3),
+                                    None, mb2, ko2, Ci, Co * kh * kw, Hp, Wp,
+                                    0, 1, int(dir3), 0)
 
         dw1 = None
         if ctx.needs_input_grad[1]:
@@ -210,7 +222,7 @@ class _GatherConvFP8Fn(torch.autograd.Function):
         mbase, koff = _plan(x.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
         bias32 = bias.float().contiguous() if bias is not None else None
         y = ext_fwd(xbuf, _wmat(w1, fp8=True), bias32, mbase, koff, Co, K,
-                    HO, WO, act, stride, 0)
+                    HO, WO, act, stride, 0, 0)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act,
                     pads, stuff, x.shape, x.dtype)
@@ -246,7 +258,7 @@ class _GatherConvFP8Fn(torch.autograd.Function):
                 mb2, ko2 = _plan(dy.device, Co, dybuf.shape[2], dybuf.shape[3],
                                  kh, kw, 1, dil, Hp, Wp)
                 dxbuf = ext_fwd(dybuf, _wmat(wrot, fp8=True), None, mb2, ko2,
-                                Ci, Co * kh * kw, Hp, Wp, 0, 1, 0)
+                                Ci, Co * kh * kw, Hp, Wp, 0, 1, 0, 0)
                 _, _, H, W = xshape
                 dx = dxbuf[:, :, pt:pt + (H - 1) * stuff + 1:stuff,
                            pl:pl + (W - 1) * stuff + 1:stuff].to(xdtype)
@@ -375,7 +387,7 @@ class _GatherConv3dFn(torch.autograd.Function):
         bias32 = bias.float().contiguous() if bias is not None else None
         # kernel sees a 2D problem: M = Do*Ho*Wo pixels, "WO" = Wo rows
         y = ext_fwd(xbuf.view(B, Ci, Dp * Hp, Wp), _wmat(w1), bias32, mbase,
-                    koff, Co, K, Do * Ho, Wo, act, 1, 0)
+                    koff, Co, K, Do * Ho, Wo, act, 1, 0, 0)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (kd, kh, kw, Do, Ho, Wo, bias is not None, act)
         return y.view(B, Co, Do, Ho, Wo)
@@ -401,7 +413,7 @@ class _GatherConv3dFn(torch.autograd.Function):
                 dxbuf = ext_fwd(dybuf.view(B, Co, -1, dybuf.shape[4]),
                                 _wmat_rot(w1, kd * kh * kw), None,
                                 mb2, ko2, Ci, Co * kd * kh * kw, Dp * Hp, Wp,
-                                0, 1, 0).view(B, Ci, Dp, Hp, Wp)
+                                0, 1, 0, 0).view(B, Ci, Dp, Hp, Wp)
 
         dw1 = None
         if ctx.needs_input_grad[1]:

@@ -23,7 +23,8 @@ std::tuple<torch::Tensor, torch::Tensor> conv_tables(
 torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        c10::optional<torch::Tensor> bias, torch::Tensor mbase,
                        torch::Tensor koff, int64_t N, int64_t K, int64_t HO,
-                       int64_t WO, int64_t act, int64_t stride, int64_t direct);
+                       int64_t WO, int64_t act, int64_t stride, int64_t direct,
+                       int64_t vpad);
 torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t WO, bool mcontig);

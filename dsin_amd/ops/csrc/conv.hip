@@ -28,7 +28,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        c10::optional<torch::Tensor> bias,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t HO, int64_t WO, int64_t act,
-                       int64_t stride, int64_t direct) {
+                       int64_t stride, int64_t direct, int64_t vpad) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(wmat);
   const bool fp8 = xbuf.scalar_type() == torch::kByte;  // raw e4m3 bytes
@@ -53,6 +53,8 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
       bp = bias->data_ptr<float>();
     }
     const int Hp = (int)xbuf.size(2), Wp = (int)xbuf.size(3);
+    TORCH_CHECK(HO == Hp + 2 * vpad - 2 && WO == Wp + 2 * vpad - 2,
+                "direct conv size mismatch");
     dim3 grid(((HO + 7) / 8) * ((WO + 7) / 8), (N + 63) / 64, B);
     size_t lds = (size_t)2 * 10 * 11 * 64 * 2;
     hipLaunchKernelGGL(conv3x3_direct_kernel, grid, dim3(256), lds,
@@ -61,7 +63,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        (const cvbf16*)wmat.data_ptr(), bp,
                        (cvbf16*)out.data_ptr(), (int)Ci, Hp, Wp, (int)N,
                        (int)HO, (int)WO, (int)KPd, xbuf.stride(0),
-                       (long long)N * M, (int)act);
+                       (long long)N * M, (int)act, (int)vpad);
     return out;
   }
   const int64_t KP = (K + 63) & ~63;  // 64-chunk padded; wmat zero-padded

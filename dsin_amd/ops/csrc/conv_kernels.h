@@ -319,6 +319,11 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
 // -- wmat_make mode 2/3.
 //   out[n][oy0+py][ox0+px] = sum_{ci,r,s} xpad[ci][oy0+py+r][ox0+px+s]
 //                                        * w[n][(ci,r,s)]
+// vp: VIRTUAL symmetric pad — when nonzero, xpad is an UNPADDED tensor
+// (Hp, Wp are its real dims) and staging clamps/zero-masks the halo
+// window instead of reading a pre-padded buffer. Used where no other
+// consumer needs the padded buffer (backward-data dy; no-grad forwards),
+// killing the pad kernel + a full tensor round trip per conv.
 __global__ __launch_bounds__(256)
 void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
                            const cvbf16* __restrict__ wmat, // (N, KP+AP)
@@ -326,7 +331,7 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
                            cvbf16* __restrict__ out,        // (N, HO, WO)
                            int Ci, int Hp, int Wp, int N, int HO, int WO,
                            int KP, long long x_img_stride,
-                           long long o_img_stride, int act) {
+                           long long o_img_stride, int act, int vp) {
   constexpr int CIC = 64;           // input channels per LDS chunk
   constexpr int XT = 11;            // LDS tile row stride in pixels (odd:
                                     // breaks 2-row bank aliasing)
@@ -361,20 +366,54 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
   };
 
   typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
-  // stage chunk c's 10x10xCIC halo tile into LDS buffer buf
+  // stage chunk c's 10x10xCIC halo tile into LDS buffer buf.
+  // vp == 0 (pre-padded source): straight vector reads; the source buffer
+  // carries >= 16 elements of tail slack (ops/conv.py) so edge overshoot
+  // is safe. vp > 0: window row/cols outside [0,Hp)x[0,Wp) are zeroed; the
+  // vector path is used only when its 40-element read provably stays
+  // inside the tensor, else a per-element guarded path runs (first/last
+  // rows of the image, left/right border tiles).
   auto stage = [&](int c, int buf) {
     char* dst = As8 + buf * TB * 2;
+    const long long nelem = (long long)Ci * Hp * Wp;
     for (int ridx = tid; ridx < CIC * 10; ridx += 256) {
       const int ci = ridx / 10, y = ridx % 10;
-      const int yy = min(oy0 + y, Hp - 1);  // bottom edge tiles: clamped
-      const cvbf16* g = x + ((long long)(c * CIC + ci) * Hp + yy) * Wp + ox0;
-      const u16x8 a = *reinterpret_cast<const u16x8*>(g);
-      const u16x8 b = *reinterpret_cast<const u16x8*>(g + 8);
+      if (vp == 0) {
+        const int yy = min(oy0 + y, Hp - 1);  // bottom edge tiles: clamped
+        const cvbf16* g = x + ((long long)(c * CIC + ci) * Hp + yy) * Wp + ox0;
+        const u16x8 a = *reinterpret_cast<const u16x8*>(g);
+        const u16x8 b = *reinterpret_cast<const u16x8*>(g + 8);
 #pragma unroll
-      for (int xi = 0; xi < 10; ++xi) {
-        unsigned short v = xi < 8 ? a[xi] : b[xi - 8];
-        *reinterpret_cast<cvbf16*>(&dst[aoff(y * XT + xi, ci)]) =
-            *reinterpret_cast<cvbf16*>(&v);
+        for (int xi = 0; xi < 10; ++xi) {
+          unsigned short v = xi < 8 ? a[xi] : b[xi - 8];
+          *reinterpret_cast<cvbf16*>(&dst[aoff(y * XT + xi, ci)]) =
+              *reinterpret_cast<cvbf16*>(&v);
+        }
+        continue;
+      }
+      const int gy = oy0 + y - vp;
+      const int cbase = ox0 - vp;
+      const bool rowin = gy >= 0 && gy < Hp;
+      const long long off =
+          ((long long)(c * CIC + ci) * Hp + (rowin ? gy : 0)) * Wp + cbase;
+      if (rowin && cbase >= 0 && cbase + 10 <= Wp && off + 40 <= nelem) {
+        const cvbf16* g = x + off;
+        const u16x8 a = *reinterpret_cast<const u16x8*>(g);
+        const u16x8 b = *reinterpret_cast<const u16x8*>(g + 8);
+#pragma unroll
+        for (int xi = 0; xi < 10; ++xi) {
+          unsigned short v = xi < 8 ? a[xi] : b[xi - 8];
+          *reinterpret_cast<cvbf16*>(&dst[aoff(y * XT + xi, ci)]) =
+              *reinterpret_cast<cvbf16*>(&v);
+        }
+      } else {
+#pragma unroll
+        for (int xi = 0; xi < 10; ++xi) {
+          cvbf16 v = cvf2b(0.f);
+          const int col = cbase + xi;
+          if (rowin && col >= 0 && col < Wp) v = x[off + xi];
+          *reinterpret_cast<cvbf16*>(&dst[aoff(y * XT + xi, ci)]) = v;
+        }
       }
     }
   };
