@@ -9,7 +9,9 @@ geometry transforms:
     buffer, flattens the weight to (Co, K=(ci,r,s)); the kernel computes
     out[pixel][cout] = sum_k xbuf[mbase[pixel]+koff[k]] * w[cout][k];
   * backward-data is ANOTHER gather conv: stride-1 dilated conv of the
-    zero-stuffed padded dy with spatially-rotated ci<->co-swapped weights;
+    zero-stuffed padded dy with spatially-rotated ci<->co-swapped weights
+    (stride-1 3x3 shapes skip the dy buffer entirely: the direct kernel
+    zero-masks the virtual halo);
   * backward-weight is the conv_wrw kernel (per-pixel-chunk fp32 partial
     slices reduced with one sum — no atomics);
   * pad / zero-stuff and the rotated W panel are single fused kernels
