@@ -70,7 +70,7 @@ def main():
         xbuf = F.pad(x.to(torch.bfloat16), (pad,) * 4).contiguous()
         mb, ko = _plan(dev, Ci, H + 2 * pad, W + 2 * pad, k, k, st, d, HO, WO)
         t_ours_w = timeit(lambda: ext_wrw(xbuf, dy, mb, ko, Co, Ci * k * k,
-                                          WO, st == 1, 0), args.iters)
+                                          WO, st == 1), args.iters)
         wg = wb.requires_grad_(True)
         def mi_w():
             y = F.conv2d(xb, wg, None, stride=st, padding=pad, dilation=d)
