@@ -34,12 +34,16 @@ def test_bench_single(tmp_path):
 
 
 def test_bench_distributed_two_ranks(tmp_path):
+    import socket
+    with socket.socket() as sock:  # pick a free rendezvous port
+        sock.bind(("127.0.0.1", 0))
+        port = sock.getsockname()[1]
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29531", os.path.join(REPO, "bench.py")] + ARGS,
+         "--master-port", str(port), os.path.join(REPO, "bench.py")] + ARGS,
         cwd=REPO, env=env, capture_output=True, text=True, timeout=900)
     assert r.returncode == 0, r.stderr[-2000:]
     lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
