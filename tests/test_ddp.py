@@ -37,7 +37,10 @@ def _worker_reducer(rank, world, port, q):
 
 
 def test_grad_reducer_averages():
-    port = 29511
+    import socket
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        port = _s.getsockname()[1]
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker_reducer, args=(r, 2, port, q))
@@ -94,7 +97,10 @@ def _worker_trainstep(rank, world, port, q):
 
 
 def test_full_train_step_replicas_stay_synced():
-    port = 29513
+    import socket
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        port = _s.getsockname()[1]
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker_trainstep, args=(r, 2, port, q))
