@@ -2,8 +2,9 @@
 
     PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
 
-Produces dsin_amd/ops/_dsin_hip*.so next to its Python dispatch layer so the
-.so travels with the repo snapshot to GPU boxes (no JIT cache dependence).
+Produces dsin_amd/ops/_dsin_hip*.so next to its Python dispatch layer. The
+.so is git-ignored (history stays source-only) but does travel with gpurun
+snapshots to GPU boxes, so there is no JIT-cache dependence at run time.
 """
 
 import glob
@@ -17,8 +18,13 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "dsin_amd", "ops", "csrc")
 
-sources = sorted(glob.glob(os.path.join(CSRC, "*.hip")) +
-                 glob.glob(os.path.join(CSRC, "*.cpp")))
+# Exclude hipify-generated "*_hip.hip" twins: torch's hipify pass recreates
+# them at build time from the hand-written sources; globbing both would
+# compile duplicate symbols (and the twins can drift from the originals).
+sources = sorted(s for s in
+                 glob.glob(os.path.join(CSRC, "*.hip")) +
+                 glob.glob(os.path.join(CSRC, "*.cpp"))
+                 if not s.endswith("_hip.hip"))
 
 setup(
     name="dsin_amd_hip",
