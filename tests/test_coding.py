@@ -60,15 +60,55 @@ def test_prediction_network_shapes(pc_config):
     assert f.shape == (6,) and (f >= 1).all()
 
 
-def test_codec_roundtrip_exact(pc_config):
-    """Bit-exact autoregressive encode/decode through the entropy model."""
+def test_codec_roundtrip(pc_config):
+    """Bit-exact wavefront encode/decode through the entropy model."""
     pc = _make_pc(pc_config)
     centers = torch.linspace(-2, 2, 6)
     torch.manual_seed(2)
     symbols = torch.randint(0, 6, (3, 5, 6))
-    data = encode_symbols(pc, centers, symbols, exact=True)
+    data = encode_symbols(pc, centers, symbols)
     out = decode_symbols(pc, centers, data, (3, 5, 6))
     assert torch.equal(out.cpu(), symbols)
+
+
+def test_codec_roundtrip_zero_padding(pc_config):
+    """use_centers_for_padding=False: encoder and decoder must agree on the
+    0.0 pad *value* (the round-1 bug paired a 0.0-value pad on encode with a
+    centers[0] symbol pad on decode)."""
+    cfg = pc_config.clone()
+    cfg.use_centers_for_padding = False
+    pc = _make_pc(cfg)
+    centers = torch.linspace(-2, 2, 6)
+    torch.manual_seed(5)
+    symbols = torch.randint(0, 6, (3, 5, 6))
+    data = encode_symbols(pc, centers, symbols)  # exact=True default
+    out = decode_symbols(pc, centers, data, (3, 5, 6))
+    assert torch.equal(out.cpu(), symbols)
+
+
+def test_wave_order_causality(pc_config):
+    """Every causal dependency of a position must land in an earlier wave."""
+    from dsin_amd.coding.entropy import _wave_order
+    waves = _wave_order(4, 7, 9, pad=4)
+    wave_of = {}
+    for i, wv in enumerate(waves):
+        for c, h, w in wv:
+            wave_of[(c, h, w)] = i
+    assert len(wave_of) == 4 * 7 * 9
+    for (c, h, w), i in wave_of.items():
+        # raster-causal in-plane deps within the +-4 window
+        for dw in range(-4, 0):
+            if (c, h, w + dw) in wave_of:
+                assert wave_of[(c, h, w + dw)] < i
+        for dh in range(-4, 0):
+            for dw in range(-4, 5):
+                if (c, h + dh, w + dw) in wave_of:
+                    assert wave_of[(c, h + dh, w + dw)] < i
+        # full previous-plane window
+        for dh in range(-4, 5):
+            for dw in range(-4, 5):
+                if (c - 1, h + dh, w + dw) in wave_of:
+                    assert wave_of[(c - 1, h + dh, w + dw)] < i
 
 
 def test_codec_bits_close_to_bitcost(pc_config):
@@ -76,7 +116,7 @@ def test_codec_bits_close_to_bitcost(pc_config):
     centers = torch.linspace(-2, 2, 6)
     torch.manual_seed(3)
     symbols = torch.randint(0, 6, (4, 8, 10))
-    data = encode_symbols(pc, centers, symbols, exact=False)
+    data = encode_symbols(pc, centers, symbols)
     est_bits = ProbclassTesting(pc, centers).total_bit_cost(symbols)
     actual_bits = len(data) * 8
     # coder overhead: freq quantization + 4 flush bytes
