@@ -103,11 +103,26 @@ class FusedAdam:
 
     @torch.no_grad()
     def step(self):
-        step_fn = _require_ext("adam_step")
         self.step_t += 1
-        step_fn(self.flat_p, self.flat_g, self.exp_avg, self.exp_avg_sq,
-                self.lr_t, self.step_t, self.wd, self.betas[0], self.betas[1],
-                self.eps)
+        if self.flat_p.is_cuda:
+            step_fn = _require_ext("adam_step")
+            step_fn(self.flat_p, self.flat_g, self.exp_avg, self.exp_avg_sq,
+                    self.lr_t, self.step_t, self.wd, self.betas[0],
+                    self.betas[1], self.eps)
+            return
+        # CPU fallback with identical flat-buffer math (TF-Adam semantics:
+        # eps outside the bias-corrected sqrt). Lets gloo CI run the SAME
+        # fused-optimizer + FlatGradReducer path the GPU uses.
+        b1, b2 = self.betas
+        t = float(self.step_t.item())
+        g = self.flat_g
+        if self.wd is not None:
+            g = g + self.wd * self.flat_p
+        self.exp_avg.mul_(b1).add_(g, alpha=1.0 - b1)
+        self.exp_avg_sq.mul_(b2).addcmul_(g, g, value=1.0 - b2)
+        lr_t = float(self.lr_t.item()) * ((1.0 - b2 ** t) ** 0.5) / (1.0 - b1 ** t)
+        denom = self.exp_avg_sq.sqrt().add_(self.eps)
+        self.flat_p.addcdiv_(self.exp_avg, denom, value=-lr_t)
 
     def state_dict(self):
         return {

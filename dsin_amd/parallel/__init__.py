@@ -1,3 +1,5 @@
-from .ddp import GradReducer, init_distributed, is_distributed, world_size, rank
+from .ddp import (FlatGradReducer, GradReducer, init_distributed,
+                  is_distributed, rank, world_size)
 
-__all__ = ["GradReducer", "init_distributed", "is_distributed", "world_size", "rank"]
+__all__ = ["FlatGradReducer", "GradReducer", "init_distributed",
+           "is_distributed", "world_size", "rank"]

@@ -23,19 +23,25 @@ from typing import Optional
 
 import torch
 
-from ..parallel import GradReducer, is_distributed
+from ..parallel import FlatGradReducer, GradReducer, is_distributed
 from .helpers import LRSchedule, create_optimizer, num_itr_per_epoch
 
 
-def _make_optimizer(config, params, device, capturable: bool):
+def _make_optimizer(config, params, device, capturable: bool,
+                    fused: Optional[bool] = None):
     kind = config.optimizer
     lr = float(config.lr_initial)
     if kind == "ADAM":
-        if device.type == "cuda":
-            from ..ops import hip_available
-            if hip_available():
-                from ..ops.adam import FusedAdam
-                return FusedAdam(params, lr=lr, betas=(0.9, 0.999), eps=1e-8)
+        use_fused = fused
+        if use_fused is None:
+            if device.type == "cuda":
+                from ..ops import hip_available
+                use_fused = hip_available()
+            else:
+                use_fused = False
+        if use_fused:
+            from ..ops.adam import FusedAdam
+            return FusedAdam(params, lr=lr, betas=(0.9, 0.999), eps=1e-8)
         if capturable:
             lr_t = torch.tensor(lr, device=device)
             return torch.optim.Adam(params, lr=lr_t, betas=(0.9, 0.999),
@@ -71,7 +77,8 @@ class Trainer:
                  use_cuda_graph: bool = False,
                  graph_warmup: int = 3,
                  ddp_bucket_bytes: int = 8 * 1024 * 1024,
-                 ddp_comm_dtype: Optional[torch.dtype] = None):
+                 ddp_comm_dtype: Optional[torch.dtype] = None,
+                 fused_adam: Optional[bool] = None):
         self.model = model
         self.ae_config = ae_config
         self.device = device or next(model.parameters()).device
@@ -81,9 +88,9 @@ class Trainer:
 
         ae_params, pc_params = model.param_groups()
         self.opt_ae = _make_optimizer(ae_config, ae_params, self.device,
-                                      self.use_cuda_graph)
+                                      self.use_cuda_graph, fused_adam)
         self.opt_pc = _make_optimizer(pc_config, pc_params, self.device,
-                                      self.use_cuda_graph)
+                                      self.use_cuda_graph, fused_adam)
         batch = ae_config.batch_size if ae_config.AE_only else 1
         itr_ep = num_itr_per_epoch(ae_config.num_crops_per_img, batch,
                                    num_training_imgs, ae_config.AE_only)
@@ -93,10 +100,17 @@ class Trainer:
 
         from ..ops.adam import FusedAdam
         self._fused = isinstance(self.opt_ae, FusedAdam)
+        self._flat_reducers = []
         if self._fused:
             self._delegate_reg_to_optimizer(model)
-            # flat gradient buffers double as the DDP communication buffers
             self.reducer = GradReducer([])  # inactive
+            if is_distributed():
+                # flat gradient buffers double as the DDP communication
+                # buffers; buckets all-reduce as backward fills them
+                self._flat_reducers = [
+                    FlatGradReducer(self.opt_ae, bucket_bytes=ddp_bucket_bytes),
+                    FlatGradReducer(self.opt_pc, bucket_bytes=ddp_bucket_bytes),
+                ]
         else:
             self.reducer = GradReducer(ae_params + pc_params,
                                        bucket_bytes=ddp_bucket_bytes,
@@ -161,26 +175,32 @@ class Trainer:
         self.opt_pc.zero_grad(set_to_none=not self.use_cuda_graph
                               or self._fused)
         self.reducer.prepare()
+        for r in self._flat_reducers:
+            r.prepare()
         with self._autocast():
             out = self.model.train_losses(x, y)
         out["loss"].backward()
-        if self._fused:
+        if self._flat_reducers:
+            # hooks gathered each bucket into flat_g and launched its
+            # all-reduce during backward; flush stragglers, wait, average
+            for r in self._flat_reducers:
+                r.finalize()
+        elif self._fused:
             # stolen per-tensor grads -> flat buffers (one fused copy)
             self.opt_ae.gather_grads()
             self.opt_pc.gather_grads()
-        if self._fused and is_distributed():
-            import torch.distributed as dist
-            ws = float(dist.get_world_size())
-            for opt in (self.opt_ae, self.opt_pc):
-                dist.all_reduce(opt.flat_g)
-                opt.flat_g.div_(ws)
-        elif not self._fused:
+        else:
             self.reducer.finalize()
         self.opt_ae.step()
         self.opt_pc.step()
         return out["loss"].detach(), out["bpp"].detach()
 
     def _try_capture(self, x, y):
+        # NOTE: capture setup performs two real warmup optimizer updates on
+        # the capture batch and the first replay trains on it once more, so
+        # that one batch is trained on three consecutive times (each counted
+        # in global_step). A one-off distribution artifact of enabling
+        # use_cuda_graph, negligible over a training run.
         try:
             self._static_x = x.clone()
             self._static_y = y.clone() if y is not None else None

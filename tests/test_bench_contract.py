@@ -33,7 +33,7 @@ def test_bench_single(tmp_path):
     _check_json(line, 1)
 
 
-def test_bench_distributed_two_ranks(tmp_path):
+def _run_distributed(n: int):
     import socket
     with socket.socket() as sock:  # pick a free rendezvous port
         sock.bind(("127.0.0.1", 0))
@@ -42,10 +42,20 @@ def test_bench_distributed_two_ranks(tmp_path):
     env["MASTER_ADDR"] = "127.0.0.1"
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--nproc-per-node", str(n), "--master-addr", "127.0.0.1",
          "--master-port", str(port), os.path.join(REPO, "bench.py")] + ARGS,
         cwd=REPO, env=env, capture_output=True, text=True, timeout=900)
     assert r.returncode == 0, r.stderr[-2000:]
     lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
     assert len(lines) == 1  # rank 0 only
-    _check_json(lines[-1], 2)
+    _check_json(lines[-1], n)
+
+
+def test_bench_distributed_two_ranks(tmp_path):
+    _run_distributed(2)
+
+
+def test_bench_distributed_four_ranks(tmp_path):
+    """The driver's scaling launch shape (torch.distributed.run, one rank
+    per GPU) at 4 ranks over gloo."""
+    _run_distributed(4)

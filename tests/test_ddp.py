@@ -74,7 +74,70 @@ def test_grad_reducer_averages():
         assert abs(torch.from_numpy(g) - e).max() < 1e-5
 
 
-def _worker_trainstep(rank, world, port, q):
+def _worker_flat_reducer(rank, world, port, q):
+    _setup(rank, world, port)
+    from dsin_amd.ops.adam import FusedAdam
+    from dsin_amd.parallel import FlatGradReducer
+    torch.manual_seed(0)  # same params on all ranks
+    model = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.ReLU(),
+                                torch.nn.Linear(32, 4))
+    params = list(model.parameters())
+    opt = FusedAdam(params, lr=0.0)  # lr 0: only the reducer matters here
+    reducer = FlatGradReducer(opt, bucket_bytes=256)  # force several buckets
+    torch.manual_seed(100 + rank)  # different data per rank
+    x = torch.randn(8, 16)
+    for p in params:
+        p.grad = None
+    reducer.prepare()
+    model(x).pow(2).mean().backward()
+    reducer.finalize()
+    q.put((rank, opt.flat_g.numpy().copy(), x.numpy(),
+           [list(s) for s in opt._slices]))
+    dist.destroy_process_group()
+
+
+def test_flat_grad_reducer_averages():
+    """The fused-path overlapped reducer (the path the GPU runs) must leave
+    flat_g equal to the cross-rank mean gradient."""
+    import socket
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        port = _s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    world = 2
+    procs = [ctx.Process(target=_worker_flat_reducer, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, flat_g, x, slices = q.get(timeout=120)
+        results[rank] = (flat_g, x, slices)
+    for p in procs:
+        p.join(timeout=60)
+
+    for r in range(1, world):
+        assert abs(results[0][0] - results[r][0]).max() < 1e-6
+
+    # equals the manual average of per-rank flat gradients
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.ReLU(),
+                                torch.nn.Linear(32, 4))
+    params = list(model.parameters())
+    slices = results[0][2]
+    accum = torch.zeros(sum(n for _, n in slices))
+    for r in range(world):
+        model.zero_grad()
+        x = torch.from_numpy(results[r][1])
+        model(x).pow(2).mean().backward()
+        for p, (ofs, n) in zip(params, slices):
+            accum[ofs:ofs + n] += p.grad.reshape(-1)
+    expect = accum / world
+    assert abs(torch.from_numpy(results[0][0]) - expect).max() < 1e-5
+
+
+def _worker_trainstep(rank, world, port, q, fused):
     _setup(rank, world, port)
     from dsin_amd import config as cm
     from dsin_amd.models import DSIN
@@ -87,31 +150,44 @@ def _worker_trainstep(rank, world, port, q):
     ae.y_patch_size = (16, 16)
     torch.manual_seed(123 + rank)  # DIFFERENT init; broadcast must fix it
     model = DSIN(ae, pc)
-    tr = Trainer(model, ae, pc, num_training_imgs=100)
+    tr = Trainer(model, ae, pc, num_training_imgs=100, fused_adam=fused)
     gen = SyntheticStereo(64, 96, seed=555 + rank)
-    x, y = gen.next_batch()
-    loss, bpp = tr.train_step(x, y)
+    for _ in range(2):
+        x, y = gen.next_batch()
+        loss, bpp = tr.train_step(x, y)
     w = model.encoder.h1.conv.weight.detach().clone()
     q.put((rank, float(loss), w.numpy()))
     dist.destroy_process_group()
 
 
-def test_full_train_step_replicas_stay_synced():
+def _run_trainstep_world(world, fused):
     import socket
     with socket.socket() as _s:
         _s.bind(("127.0.0.1", 0))
         port = _s.getsockname()[1]
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker_trainstep, args=(r, 2, port, q))
-             for r in range(2)]
+    procs = [ctx.Process(target=_worker_trainstep,
+                         args=(r, world, port, q, fused))
+             for r in range(world)]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(2):
-        rank, loss, w = q.get(timeout=300)
+    for _ in range(world):
+        rank, loss, w = q.get(timeout=600)
         results[rank] = (loss, w)
     for p in procs:
         p.join(timeout=60)
     # identical post-step weights despite different data and different seeds
-    assert abs(results[0][1] - results[1][1]).max() < 1e-6
+    for r in range(1, world):
+        assert abs(results[0][1] - results[r][1]).max() < 1e-6
+
+
+def test_full_train_step_replicas_stay_synced():
+    _run_trainstep_world(2, fused=None)
+
+
+def test_full_train_step_fused_flat_reducer_world4():
+    """4-rank fused-Adam + FlatGradReducer — the exact optimizer/reducer
+    combination an 8-GPU RCCL run uses, over gloo on CPU."""
+    _run_trainstep_world(4, fused=True)
