@@ -215,12 +215,22 @@ class SyntheticDataset:
 
 
 def make_dataset(config, data_paths_dir: str, seed: int = 0):
-    """Dataset factory: path-pair lists when present, synthetic otherwise."""
+    """Dataset factory: the file-backed Dataset when the lists AND the
+    images they reference are present, synthetic otherwise (stated loudly).
+    """
+    import warnings
     if str(config.root_data) == "synthetic":
         return SyntheticDataset(config)
     train_list = os.path.join(data_paths_dir, config.file_path_train)
     if not os.path.exists(train_list):
-        import warnings
         warnings.warn(f"{train_list} not found - using synthetic data")
+        return SyntheticDataset(config)
+    pairs = read_pair_list(train_list, config.root_data)
+    if not pairs or not os.path.exists(pairs[0][0]):
+        probe = pairs[0][0] if pairs else train_list
+        warnings.warn(
+            f"data list {train_list} present but images missing "
+            f"(probed {probe}); set root_data to the KITTI root to train "
+            "on real data - using synthetic data")
         return SyntheticDataset(config)
     return Dataset(config, data_paths_dir, seed=seed)

@@ -85,6 +85,25 @@ def loss_list_saver(x: np.ndarray, y: np.ndarray, x_rec: np.ndarray,
         appendf("pearson", pearson_per_patch(x0, y_syn[0], ph, pw))
 
 
+def eval_msssim_bpp(model, x, y):
+    """MS-SSIM (numpy 5-scale oracle) + bpp of the model's reconstruction of
+    one batch — the quality half of the headline metric ("MS-SSIM @ 0.02
+    bpp"). Uses x_with_si when side information is active,
+    x_dec otherwise. Returns (msssim, bpp) floats; msssim is NaN when the
+    crop is too small for 5 dyadic scales (min side < 176)."""
+    import torch
+    with torch.no_grad():
+        _, _, x_dec, x_with_si, bpp = model.reconstruct(x, y)
+        x_rec = (x_with_si if x_with_si is not None and
+                 float(x_with_si.abs().mean()) > 0 else x_dec)
+        x_np = x.float().cpu().numpy()
+        r_np = x_rec.float().clamp(0, 255).cpu().numpy()
+    if min(x_np.shape[-2], x_np.shape[-1]) < 176:
+        return float("nan"), float(bpp)
+    nhwc = lambda img: np.transpose(img, (0, 2, 3, 1))
+    return float(multiscale_ssim_np(nhwc(x_np), nhwc(r_np))), float(bpp)
+
+
 class MetricsLogger:
     """JSONL metrics stream: one line per event, flushed immediately."""
 

@@ -20,7 +20,8 @@ from dsin_amd.data import make_dataset
 from dsin_amd.models import DSIN
 from dsin_amd.parallel import init_distributed, rank
 from dsin_amd.training import Trainer, checkpoint
-from dsin_amd.utils import MetricsLogger, loss_list_saver, save_test_img
+from dsin_amd.utils import (MetricsLogger, eval_msssim_bpp, loss_list_saver,
+                            save_test_img)
 
 
 def get_validate_every(iteration, total_iterations, validate_every, p1, p2):
@@ -49,6 +50,10 @@ def main(argv=None):
                     help="periodic crash-recovery checkpoint interval "
                          "(iterations; 0 = best-val only, the reference's "
                          "behavior)")
+    ap.add_argument("--plots", action="store_true",
+                    help="save loss / inference figures (headless "
+                         "equivalents of the reference's plt.show windows, "
+                         "src/utils.py:12-79) under the images root")
     args = ap.parse_args(argv)
 
     ae_config, _ = config_mod.parse(args.ae_config_path)
@@ -89,6 +94,9 @@ def main(argv=None):
         val_iterations = max(len(val_names) // data.batch_size, 1)
         train_sum = bpp_sum = 0.0
         show_every = int(ae_config.show_every)
+        train_hist, train_hist_iters = [], []
+        val_hist, val_hist_iters = [], []
+        best_iter = 0
         for iteration in range(1, total_iterations + 1):
             x, y = data.get_data_for_train()
             x, y = x.to(device), y.to(device)
@@ -104,11 +112,20 @@ def main(argv=None):
                 val_sum = 0.0
                 for _ in range(val_iterations):
                     xv, yv = data.get_data_for_val()
-                    val_sum += float(trainer.validate(xv.to(device), yv.to(device)))
+                    xv, yv = xv.to(device), yv.to(device)
+                    val_sum += float(trainer.validate(xv, yv))
                 val_loss = val_sum / val_iterations
-                metrics.log("val", iteration=iteration, val_loss=val_loss)
+                # RD operating point: MS-SSIM (numpy oracle) + bpp on the
+                # last val batch — the quality half of the headline metric
+                # (BASELINE: "imgs/sec ...; MS-SSIM @ 0.02 bpp")
+                val_ms, val_bpp = eval_msssim_bpp(model, xv, yv)
+                val_hist.append(val_loss)
+                val_hist_iters.append(iteration)
+                metrics.log("val", iteration=iteration, val_loss=val_loss,
+                            val_msssim=val_ms, val_bpp=val_bpp)
                 if val_loss < best_val and rank() == 0:
                     best_val = val_loss
+                    best_iter = iteration
                     if ae_config.save_model:
                         model_name = checkpoint.model_name_for(ae_config, now)
                         checkpoint.save(model, trainer.optimizers, trainer.global_step,
@@ -128,7 +145,18 @@ def main(argv=None):
                       f"bpp {bpp_sum/show_every:.4f} val {val_loss:.4f}")
                 metrics.log("train", iteration=iteration,
                             loss=train_sum / show_every, bpp=bpp_sum / show_every)
+                train_hist.append(train_sum / show_every)
+                train_hist_iters.append(iteration)
                 train_sum = bpp_sum = 0.0
+
+        if args.plots and rank() == 0 and train_hist:
+            from dsin_amd.utils.plots import plot_loss
+            path = plot_loss(train_hist, val_hist, val_hist_iters,
+                             train_hist_iters, total_iterations, best_val,
+                             best_iter, str(model_name),
+                             out_path=os.path.join(root_save_img,
+                                                   f"loss_{model_name}.png"))
+            print(f"Saved loss figure {path}")
 
     if ae_config.test_model and rank() == 0:
         _, test_names = data.get_data_size()
@@ -144,6 +172,17 @@ def main(argv=None):
                             y_syn.cpu().numpy() if y_syn is not None else None,
                             str(model_name), float(bpp), root_save_img,
                             *(int(v) for v in ae_config.y_patch_size))
+            if args.plots and i == 0 and y_syn is not None:
+                from dsin_amd.utils.plots import plot_inference
+                path = plot_inference(
+                    x[0].cpu().numpy(), x_dec[0].float().cpu().numpy(),
+                    y[0].cpu().numpy(), y_syn[0].float().cpu().numpy(),
+                    x_with_si[0].float().cpu().numpy(), str(model_name),
+                    total_iterations, cnt=trainer.global_step,
+                    bpp=f"{float(bpp):.4f}",
+                    out_path=os.path.join(root_save_img,
+                                          f"inference_{model_name}.png"))
+                print(f"Saved inference figure {path}")
             print(f"test image {i}: bpp {float(bpp):.4f}")
 
     metrics.close()
