@@ -177,6 +177,8 @@ class DSIN(nn.Module):
     def reconstruct(self, x: torch.Tensor, y: Optional[torch.Tensor]):
         """Test-time reconstruction (reference siNet_get_reconstructed,
         src/AE.py:132-148): returns (y_dec, y_syn, x_dec, x_with_si, bpp)."""
+        from ..ops import conv as _conv
+        _conv.begin_step()  # weights may have stepped since panels were built
         self.eval()
         z, x_dec = self.autoencode(x)
         bc = self.probclass.bitcost(z.qbar.float(), z.symbols, self._pad_value())

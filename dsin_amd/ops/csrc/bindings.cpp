@@ -20,14 +20,19 @@ torch::Tensor mfma32_selftest(torch::Tensor A, torch::Tensor B);
 std::tuple<torch::Tensor, torch::Tensor> conv_tables(
     int64_t M, int64_t K, int64_t WO, int64_t stride, int64_t dil, int64_t Wp,
     int64_t HpWp, int64_t kh, int64_t kw, torch::Device device);
+std::tuple<torch::Tensor, torch::Tensor> conv_tables_v2(
+    int64_t M, int64_t K, int64_t WO, int64_t stride, int64_t dil, int64_t kh,
+    int64_t kw, torch::Device device);
 torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        c10::optional<torch::Tensor> bias, torch::Tensor mbase,
                        torch::Tensor koff, int64_t N, int64_t K, int64_t HO,
                        int64_t WO, int64_t act, int64_t stride, int64_t direct,
-                       int64_t vpad);
+                       int64_t vpad, int64_t vm, int64_t vpt, int64_t vpl,
+                       int64_t vsv);
 torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
-                       int64_t K, int64_t WO, bool mcontig);
+                       int64_t K, int64_t WO, bool mcontig, int64_t vm,
+                       int64_t st, int64_t vpt, int64_t vpl, int64_t vsv);
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act);
 torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, int64_t mode);
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
@@ -56,6 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_selftest", &dsin::mfma_selftest, "MFMA 16x16x32 layout check");
   m.def("mfma32_selftest", &dsin::mfma32_selftest, "MFMA 32x32x16 layout check");
   m.def("conv_tables", &dsin::conv_tables, "gather-conv offset tables");
+  m.def("conv_tables_v2", &dsin::conv_tables_v2,
+        "packed-coordinate tables for the virtual-pad gather path");
   m.def("conv_fwd", &dsin::conv_fwd, "implicit-GEMM gather conv forward");
   m.def("conv_wrw", &dsin::conv_wrw, "implicit-GEMM conv weight gradient");
   m.def("act_bwd", &dsin::act_bwd, "fused activation gradient (bf16 out)");

@@ -1,8 +1,9 @@
+#include "hip/hip_runtime.h"
 // Host wrappers for the implicit-GEMM conv family (device code in
 // conv_kernels.h). See dsin_amd/ops/conv.py for the autograd layer and the
 // geometry-to-table mapping.
 
-#include "common.h"
+#include "common_hip.h"
 #include "conv_kernels.h"
 #define DSIN_CONV_FP8_KERNELS
 #include "conv_fp8.h"
@@ -17,7 +18,7 @@ std::tuple<torch::Tensor, torch::Tensor> conv_tables(
   auto koff = torch::empty({K}, opts);
   int64_t n = std::max(M, K);
   hipLaunchKernelGGL(conv_tables_kernel, grid1d(n, 256), dim3(256), 0,
-                     at::cuda::getCurrentCUDAStream(),
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                      mbase.data_ptr<int>(), koff.data_ptr<int>(), (int)M,
                      (int)K, (int)WO, (int)stride, (int)dil, (int)Wp,
                      (int)HpWp, (int)(kh * kw), (int)kw);
@@ -32,7 +33,7 @@ std::tuple<torch::Tensor, torch::Tensor> conv_tables_v2(
   auto kpack = torch::empty({K}, opts);
   int64_t n = std::max(M, K);
   hipLaunchKernelGGL(conv_tables_v2_kernel, grid1d(n, 256), dim3(256), 0,
-                     at::cuda::getCurrentCUDAStream(), mpack.data_ptr<int>(),
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(), mpack.data_ptr<int>(),
                      kpack.data_ptr<int>(), (int)M, (int)K, (int)WO,
                      (int)stride, (int)dil, (int)(kh * kw), (int)kw);
   return {mpack, kpack};
@@ -73,7 +74,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
     dim3 grid(((HO + 7) / 8) * ((WO + 7) / 8), (N + 63) / 64, B);
     size_t lds = (size_t)2 * 10 * 11 * 64 * 2;
     hipLaunchKernelGGL(conv3x3_direct_kernel, grid, dim3(256), lds,
-                       at::cuda::getCurrentCUDAStream(),
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                        (const cvbf16*)xbuf.data_ptr(),
                        (const cvbf16*)wmat.data_ptr(), bp,
                        (cvbf16*)out.data_ptr(), (int)Ci, Hp, Wp, (int)N,
@@ -95,7 +96,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   if (fp8) {
     size_t lds = (size_t)2 * CONV8_TM * (64 + CONV8_AP);
     hipLaunchKernelGGL(conv_fwd_fp8_kernel, grid, dim3(256), lds,
-                       at::cuda::getCurrentCUDAStream(),
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                        (const f8*)xbuf.data_ptr(), (const f8*)wmat.data_ptr(),
                        bptr, (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
                        koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
@@ -113,7 +114,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
     size_t lds32 = (size_t)4 * 32 * (64 + CONV_AP) * 2;
     auto kern = vm ? conv_fwd_kernel<32, 1> : conv_fwd_kernel<32, 0>;
     hipLaunchKernelGGL(kern, grid32, dim3(256), lds32,
-                       at::cuda::getCurrentCUDAStream(),
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                        (const cvbf16*)xbuf.data_ptr(),
                        (const cvbf16*)wmat.data_ptr(), bptr,
                        (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
@@ -125,7 +126,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   size_t lds = (size_t)4 * CONV_TM * (64 + CONV_AP) * 2;  // 4-buffer pipeline
   auto kern = vm ? conv_fwd_kernel<64, 1> : conv_fwd_kernel<64, 0>;
   hipLaunchKernelGGL(kern, grid, dim3(256), lds,
-                     at::cuda::getCurrentCUDAStream(),
+                     at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                      (const cvbf16*)xbuf.data_ptr(),
                      (const cvbf16*)wmat.data_ptr(), bptr,
                      (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
@@ -164,7 +165,7 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
     TORCH_CHECK(dy.scalar_type() == torch::kByte, "fp8 wrw: dy must be e4m3");
     size_t lds = (size_t)4 * 64 * (32 + CONV8_AP);
     hipLaunchKernelGGL(conv_wrw_fp8_kernel, grid, dim3(256), lds,
-                       at::cuda::getCurrentCUDAStream(),
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                        (const f8*)xbuf.data_ptr(), (const f8*)dy.data_ptr(),
                        dwp.data_ptr<float>(), mbase.data_ptr<int>(),
                        koff.data_ptr<int>(), (int)M, (int)N, (int)K,
@@ -175,7 +176,7 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
     const int vH = (int)xbuf.size(2), vW = (int)xbuf.size(3);
     auto kern = vm ? conv_wrw_kernel<1> : conv_wrw_kernel<0>;
     hipLaunchKernelGGL(kern, grid, dim3(256), lds,
-                       at::cuda::getCurrentCUDAStream(),
+                       at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                        (const cvbf16*)xbuf.data_ptr(),
                        (const cvbf16*)dy.data_ptr(), dwp.data_ptr<float>(),
                        mbase.data_ptr<int>(), koff.data_ptr<int>(), (int)M,
@@ -192,7 +193,7 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act) {
   CHECK_CUDA_CONTIG(y);
   const long long n = dy.numel();
   auto out = torch::empty_like(y);  // y is bf16, same shape
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const int grid = (int)std::min<long long>((n + 255) / 256, 4096);
   if (dy.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((act_bwd_kernel<float>), dim3(grid), dim3(256), 0,
@@ -220,7 +221,7 @@ torch::Tensor wmat_make(torch::Tensor w1, int64_t khw, int64_t mode) {
                           w1.options().dtype(torch::kBFloat16));
   const long long total = rows * KPA;
   const int grid = (int)std::min<long long>((total + 255) / 256, 4096);
-  auto stream = at::cuda::getCurrentCUDAStream();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   if (w1.scalar_type() == torch::kFloat32) {
     hipLaunchKernelGGL((wmat_make_kernel<float>), dim3(grid), dim3(256), 0,
                        stream, w1.data_ptr<float>(), (cvbf16*)out.data_ptr(),

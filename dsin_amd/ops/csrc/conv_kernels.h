@@ -61,14 +61,110 @@ __global__ void conv_tables_kernel(int* __restrict__ mbase,
   }
 }
 
+// Packed-coordinate tables for the VIRTUAL-pad gather path (VM=1): instead
+// of flat offsets into a pre-padded buffer, each entry carries 2D (or
+// channel+2D) coordinates so the staging can bounds-test against the REAL
+// tensor and produce zeros for the pad/stuff positions — no padded buffer,
+// no pad_stuff kernel, no extra HBM round trip per conv.
+//   mpack[m] = (oh*stride) << 16 | (ow*stride)        (virtual-image coords)
+//   kpack[k] = ci << 20 | (r*dil) << 10 | (s*dil)
+// Field limits: ow*stride < 65536, ci < 4096, tap reach < 1024 — all DSIN
+// geometries (incl. 1024x2048 fp8 crops and siNet dil=128) fit.
+__global__ void conv_tables_v2_kernel(int* __restrict__ mpack,
+                                      int* __restrict__ kpack,
+                                      int M, int K, int WO, int stride,
+                                      int dil, int khw, int kw) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < M) {
+    int oh = i / WO, ow = i % WO;
+    mpack[i] = ((oh * stride) << 16) | (ow * stride);
+  }
+  if (i < K) {
+    int ci = i / khw, rem = i % khw;
+    int r = rem / kw, s = rem % kw;
+    kpack[i] = (ci << 20) | ((r * dil) << 10) | (s * dil);
+  }
+}
+
+typedef __attribute__((ext_vector_type(8))) unsigned short cv_u16x8;
+typedef __attribute__((ext_vector_type(4))) unsigned short cv_u16x4;
+
+// Stage one 8-element run from the virtual image pad(stuff(x, SV)) at
+// virtual coords (hv, wv0 + i*ST), i = 0..7 (pads already subtracted by the
+// caller). Elements outside the real tensor (pad region, stuff zeros,
+// borders) come out 0. Vector reads only where provably in-row; the guarded
+// per-element path touches only valid addresses, so no slack allocation or
+// address clamping is needed.
+template <int ST, int SV>
+__device__ __forceinline__ void vstage8(const cvbf16* __restrict__ x,
+                                        int H, int W, int ci, int hv,
+                                        int wv0, cv_u16x8& out) {
+  if (SV == 1) {
+    const bool rok = (unsigned)hv < (unsigned)H;
+    const long long row = ((long long)ci * H + (rok ? hv : 0)) * (long long)W;
+    if (rok && wv0 >= 0 && wv0 + (ST == 1 ? 8 : 16) <= W) {
+      if (ST == 1) {
+        out = *reinterpret_cast<const cv_u16x8*>(&x[row + wv0]);
+      } else {  // ST == 2: two contiguous reads, pick every other element
+        const cv_u16x8 a0 = *reinterpret_cast<const cv_u16x8*>(&x[row + wv0]);
+        const cv_u16x8 a1 =
+            *reinterpret_cast<const cv_u16x8*>(&x[row + wv0 + 8]);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          out[i] = a0[2 * i];
+          out[4 + i] = a1[2 * i];
+        }
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int c = wv0 + i * ST;
+        cvbf16 v = cvf2b(0.f);
+        if (rok && (unsigned)c < (unsigned)W) v = x[row + c];
+        out[i] = *reinterpret_cast<unsigned short*>(&v);
+      }
+    }
+  } else {  // SV == 2 (zero-stuffed input), ST == 1 always in this regime
+    const bool rok = hv >= 0 && !(hv & 1) && (hv >> 1) < H;
+    const long long row =
+        ((long long)ci * H + (rok ? (hv >> 1) : 0)) * (long long)W;
+    const int par = wv0 & 1;          // valid i have (wv0 + i) even
+    const int c0 = (wv0 + par) >> 1;  // real col of the first valid element
+    if (rok && c0 >= 0 && c0 + 4 <= W) {
+      const cv_u16x4 v = *reinterpret_cast<const cv_u16x4*>(&x[row + c0]);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const unsigned short e =
+            par ? (i >= 1 ? v[(i - 1) >> 1] : (unsigned short)0)
+                : v[i >> 1];
+        out[i] = ((i & 1) == par) ? e : (unsigned short)0;
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int c = wv0 + i;
+        cvbf16 v = cvf2b(0.f);
+        if (rok && !(c & 1) && (unsigned)(c >> 1) < (unsigned)W)
+          v = x[row + (c >> 1)];
+        out[i] = *reinterpret_cast<unsigned short*>(&v);
+      }
+    }
+  }
+}
+
 // TM = output pixels per workgroup. 64 is the throughput tile; 32 doubles
 // the workgroup count for small-M layers (e.g. the 40x120 resblock convs,
 // which at TM=64 launch only ~600 workgroups on 256 CUs and run
 // latency-bound at ~2.3 waves/SIMD). The host picks 32 when the TM=64 grid
 // would underfill the chip.
-template <int TM>
+// VM=0: flat-offset tables over a pre-padded buffer (legacy; conv3d).
+// VM=1: packed-coordinate tables + VIRTUAL pad/stuff — xpad is the RAW
+// (Ci, vH, vW) tensor and staging zero-masks the pad ring and stuff holes
+// (vpt/vpl = virtual pads, vsv = stuff stride). Kills the pad_stuff kernel
+// and the padded-buffer HBM round trip per conv.
+template <int TM, int VM>
 __global__ __launch_bounds__(256)
-void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
+void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // see VM note above
                      const cvbf16* __restrict__ wmat,   // (Co, KP64+AP) 0-pad
                      const float* __restrict__ bias,    // (Co,) or nullptr
                      cvbf16* __restrict__ out,          // (Co, M) i.e. NCHW
@@ -78,7 +174,7 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
                      long long x_img_stride,            // Ci*Hp*Wp
                      long long o_img_stride,            // Co*M
                      int act, int WO,
-                     int stride) {
+                     int stride, int vH, int vW, int vpt, int vpl, int vsv) {
   // Branchless pipeline: K chunks of 64 (two MFMA k-steps per barrier),
   // double-buffered LDS A-tile with XOR-swizzled addressing. Staging is two
   // overlapped 16B loads per 8-pixel run with a per-element crossing select
@@ -118,11 +214,19 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   const int sk = tid / PPT;                // 0..KCOV-1
   const int gm0 = min(m0 + sm8, M - 1);
   const int gm7 = min(m0 + sm8 + 7, M - 1);
-  const int mb0 = mbase[gm0];
-  const int mb1 = mbase[gm7] - 7 * stride; // base so elem i sits at +i*stride
+  const int mbv0 = mbase[gm0];
+  const int mbv1 = mbase[gm7];
+  // VM=0: flat base offsets; elem i sits at +i*stride from the run base.
+  const int mb0 = VM ? 0 : mbv0;
+  const int mb1 = VM ? 0 : mbv1 - 7 * stride;
+  // VM=1: virtual coords of the two runs (A: leading row, B: trailing row)
+  const int mhA = VM ? (mbv0 >> 16) : 0;
+  const int mwA = VM ? (mbv0 & 0xffff) : 0;
+  const int mhB = VM ? (mbv1 >> 16) : 0;
+  const int mwB = VM ? ((mbv1 & 0xffff) - 7 * stride) : 0;
   // crossing point: first i whose pixel falls on the next output row
   const int cross = WO - (gm0 % WO);       // >= 8 means no crossing
-  // xbuf carries >= 16 elements of tail slack (ops/conv.py _padded_buf), so
+  // VM=0: xbuf carries >= 16 elements of tail slack (ops/conv.py), so
   // in-range vector reads are never clamped; only negative bases are.
 
   const int ncol = n0 + colL;
@@ -133,6 +237,46 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // (Ci, Hp, Wp), padded
   cv_bf16x8 wfrag[2];
 
   auto load_half = [&](int ko, u16x8& st) {
+    if (VM) {
+      const int kci = ko >> 20;
+      const int kdh = (ko >> 10) & 1023;
+      const int kdw = ko & 1023;
+      if (stride == 1 || stride == 2) {
+        cv_u16x8 sa, sb;
+        const int hA = mhA + kdh - vpt, wA = mwA + kdw - vpl;
+        const int hB = mhB + kdh - vpt, wB = mwB + kdw - vpl;
+        if (vsv == 2) {
+          vstage8<1, 2>(x, vH, vW, kci, hA, wA, sa);
+          vstage8<1, 2>(x, vH, vW, kci, hB, wB, sb);
+        } else if (stride == 1) {
+          vstage8<1, 1>(x, vH, vW, kci, hA, wA, sa);
+          vstage8<1, 1>(x, vH, vW, kci, hB, wB, sb);
+        } else {
+          vstage8<2, 1>(x, vH, vW, kci, hA, wA, sa);
+          vstage8<2, 1>(x, vH, vW, kci, hB, wB, sb);
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) st[i] = (i < cross) ? sa[i] : sb[i];
+      } else {  // generic (narrow WO): per-element decode + guarded load
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          const int mp = mbase[min(m0 + sm8 + i, M - 1)];
+          const int hv = (mp >> 16) + kdh - vpt;
+          const int wv = (mp & 0xffff) + kdw - vpl;
+          cvbf16 v = cvf2b(0.f);
+          if (vsv == 2) {
+            if (hv >= 0 && !(hv & 1) && (hv >> 1) < vH && wv >= 0 &&
+                !(wv & 1) && (wv >> 1) < vW)
+              v = x[((long long)kci * vH + (hv >> 1)) * vW + (wv >> 1)];
+          } else {
+            if ((unsigned)hv < (unsigned)vH && (unsigned)wv < (unsigned)vW)
+              v = x[((long long)kci * vH + hv) * vW + wv];
+          }
+          st[i] = *reinterpret_cast<unsigned short*>(&v);
+        }
+      }
+      return;
+    }
     if (stride == 1) {
       const u16x8 a = *reinterpret_cast<const u16x8*>(&x[mb0 + ko]);
       const u16x8 b = *reinterpret_cast<const u16x8*>(
@@ -504,6 +648,7 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
 // dW[co][k] += sum over the workgroup's pixel chunk of dy[co][m]*A[m][k].
 // GEMM roles: A' (M'=filter taps K) gathered rows, B' = dy columns.
 // Tile: M'64 (taps) x N'64 (couts), K' = pixels chunked by 32.
+template <int VM>
 __global__ __launch_bounds__(256)
 void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
                      const cvbf16* __restrict__ dy,    // (Co, M)
@@ -512,7 +657,8 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
                      const int* __restrict__ koff,
                      int M, int N, int K,
                      long long x_img_stride, long long dy_img_stride,
-                     int pix_chunks, int WO, int mcontig) {
+                     int pix_chunks, int WO, int mcontig,
+                     int st, int vH, int vW, int vpt, int vpl, int vsv) {
   // blockIdx.x: tap tile; blockIdx.y: cout tile; blockIdx.z: pixel chunk*img
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int ASTR = 32 + CONV_AP;
@@ -552,12 +698,65 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
 
   typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
   u16x8 sa, sb;
+  // VM=1: the tap's virtual-coordinate deltas, decoded once
+  const int kci = VM ? (ko >> 20) : 0;
+  const int kdh = VM ? ((ko >> 10) & 1023) : 0;
+  const int kdw = VM ? (ko & 1023) : 0;
 
   auto load_chunk = [&](int pp) {
     const int p = pp + sp8;
     const bool inb = p + 7 < p1;
-    const bool avec = inb && mcontig && ((p % WO) + 8 <= WO);
-    if (avec && tap < K) {
+    if (VM) {
+      if (tap < K && p < p1) {
+        const int gm0 = min(p, M - 1), gm7 = min(p + 7, M - 1);
+        const int mpA = mbase[gm0], mpB = mbase[gm7];
+        const bool gen = (WO < 8) || (st != 1 && st != 2);
+        if (!gen) {
+          const int hA = (mpA >> 16) + kdh - vpt;
+          const int wA = (mpA & 0xffff) + kdw - vpl;
+          const int hB = (mpB >> 16) + kdh - vpt;
+          const int wB = (mpB & 0xffff) - 7 * st + kdw - vpl;
+          const int cross = WO - (gm0 % WO);
+          cv_u16x8 va, vb;
+          if (vsv == 2) {
+            vstage8<1, 2>(x, vH, vW, kci, hA, wA, va);
+            vstage8<1, 2>(x, vH, vW, kci, hB, wB, vb);
+          } else if (st == 1) {
+            vstage8<1, 1>(x, vH, vW, kci, hA, wA, va);
+            vstage8<1, 1>(x, vH, vW, kci, hB, wB, vb);
+          } else {
+            vstage8<2, 1>(x, vH, vW, kci, hA, wA, va);
+            vstage8<2, 1>(x, vH, vW, kci, hB, wB, vb);
+          }
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            sa[i] = (p + i < p1) ? ((i < cross) ? va[i] : vb[i])
+                                 : (unsigned short)0;
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            cvbf16 v = cvf2b(0.f);
+            if (p + i < p1) {
+              const int mp = mbase[min(p + i, M - 1)];
+              const int hv = (mp >> 16) + kdh - vpt;
+              const int wv = (mp & 0xffff) + kdw - vpl;
+              if (vsv == 2) {
+                if (hv >= 0 && !(hv & 1) && (hv >> 1) < vH && wv >= 0 &&
+                    !(wv & 1) && (wv >> 1) < vW)
+                  v = x[((long long)kci * vH + (hv >> 1)) * vW + (wv >> 1)];
+              } else {
+                if ((unsigned)hv < (unsigned)vH && (unsigned)wv < (unsigned)vW)
+                  v = x[((long long)kci * vH + hv) * vW + wv];
+              }
+            }
+            sa[i] = *reinterpret_cast<unsigned short*>(&v);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) sa[i] = 0;
+      }
+    } else if (inb && mcontig && ((p % WO) + 8 <= WO) && tap < K) {
       sa = *reinterpret_cast<const u16x8*>(&x[mbase[p] + ko]);
     } else {
 #pragma unroll

@@ -225,6 +225,8 @@ class Trainer:
 
     def train_step(self, x: torch.Tensor, y: Optional[torch.Tensor]):
         """Returns (loss, bpp) tensors on device (no host sync)."""
+        from ..ops import conv as _conv
+        _conv.begin_step()  # weights changed since the last step's optimizer
         self.sched_ae.set_step(self.global_step)
         self.sched_pc.set_step(self.global_step)
 
@@ -254,5 +256,7 @@ class Trainer:
         return out
 
     def validate(self, x: torch.Tensor, y: Optional[torch.Tensor]) -> torch.Tensor:
+        from ..ops import conv as _conv
+        _conv.begin_step()  # weights may have stepped since panels were built
         with self._autocast():
             return self.model.validate_loss(x, y)
