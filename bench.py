@@ -129,6 +129,13 @@ def main():
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
+    # which execution path actually ran (per-rank proof)
+    import sys
+    graph_live = getattr(trainer, "_graph", None) is not None
+    print(f"[bench rank {rank()}] path={'hipGraph' if graph_live else 'eager'}"
+          f" fused_adam={trainer._fused}"
+          f" graph_failed={trainer._graph_failed}", file=sys.stderr)
+
     # max over ranks
     if is_distributed():
         t = torch.tensor([elapsed], device=device if device.type == "cuda" else None)

@@ -108,12 +108,34 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   // (< ~3 workgroups/CU) halve the pixel tile to double parallelism — the
   // small-M resblock convs (40x120) are latency-bound, not FLOP-bound.
   const int vH = (int)xbuf.size(2), vW = (int)xbuf.size(3);
+  using FwdKern = void (*)(const cvbf16*, const cvbf16*, const float*,
+                           cvbf16*, const int*, const int*, int, int, int,
+                           int, long long, long long, int, int, int, int,
+                           int, int, int, int);
+  auto pick_fwd = [&](bool tm32) -> FwdKern {
+    if (!vm) return tm32 ? conv_fwd_kernel<32, 0, 0, 1>
+                         : conv_fwd_kernel<64, 0, 0, 1>;
+    // compile-time (stride, stuff) specialization for the virtual path
+    if (vsv == 2) {
+      if (stride == 1)
+        return tm32 ? conv_fwd_kernel<32, 1, 1, 2>
+                    : conv_fwd_kernel<64, 1, 1, 2>;
+      return tm32 ? conv_fwd_kernel<32, 1, 0, 2>
+                  : conv_fwd_kernel<64, 1, 0, 2>;
+    }
+    if (stride == 1)
+      return tm32 ? conv_fwd_kernel<32, 1, 1, 1>
+                  : conv_fwd_kernel<64, 1, 1, 1>;
+    if (stride == 2)
+      return tm32 ? conv_fwd_kernel<32, 1, 2, 1>
+                  : conv_fwd_kernel<64, 1, 2, 1>;
+    return tm32 ? conv_fwd_kernel<32, 1, 0, 1> : conv_fwd_kernel<64, 1, 0, 1>;
+  };
   const long long wgs64 = (long long)grid.x * grid.y * grid.z;
   if (wgs64 < 256 && M > CONV_TM) {
     dim3 grid32((M + 31) / 32, grid.y, grid.z);
     size_t lds32 = (size_t)4 * 32 * (64 + CONV_AP) * 2;
-    auto kern = vm ? conv_fwd_kernel<32, 1> : conv_fwd_kernel<32, 0>;
-    hipLaunchKernelGGL(kern, grid32, dim3(256), lds32,
+    hipLaunchKernelGGL(pick_fwd(true), grid32, dim3(256), lds32,
                        at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                        (const cvbf16*)xbuf.data_ptr(),
                        (const cvbf16*)wmat.data_ptr(), bptr,
@@ -124,8 +146,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
     return out;
   }
   size_t lds = (size_t)4 * CONV_TM * (64 + CONV_AP) * 2;  // 4-buffer pipeline
-  auto kern = vm ? conv_fwd_kernel<64, 1> : conv_fwd_kernel<64, 0>;
-  hipLaunchKernelGGL(kern, grid, dim3(256), lds,
+  hipLaunchKernelGGL(pick_fwd(false), grid, dim3(256), lds,
                      at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                      (const cvbf16*)xbuf.data_ptr(),
                      (const cvbf16*)wmat.data_ptr(), bptr,
@@ -174,7 +195,23 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
   } else {
     size_t lds = (size_t)4 * 64 * (32 + CONV_AP) * 2;  // 2 tiles x dbuf
     const int vH = (int)xbuf.size(2), vW = (int)xbuf.size(3);
-    auto kern = vm ? conv_wrw_kernel<1> : conv_wrw_kernel<0>;
+    using WrwKern = void (*)(const cvbf16*, const cvbf16*, float*,
+                             const int*, const int*, int, int, int,
+                             long long, long long, int, int, int, int, int,
+                             int, int, int, int);
+    WrwKern kern;
+    if (!vm) {
+      kern = conv_wrw_kernel<0, 0, 1>;
+    } else {
+      const bool gen = (WO < 8) || (st != 1 && st != 2);
+      if (vsv == 2)
+        kern = (!gen && st == 1) ? conv_wrw_kernel<1, 1, 2>
+                                 : conv_wrw_kernel<1, 0, 2>;
+      else if (gen)
+        kern = conv_wrw_kernel<1, 0, 1>;
+      else
+        kern = (st == 1) ? conv_wrw_kernel<1, 1, 1> : conv_wrw_kernel<1, 2, 1>;
+    }
     hipLaunchKernelGGL(kern, grid, dim3(256), lds,
                        at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
                        (const cvbf16*)xbuf.data_ptr(),

@@ -95,6 +95,12 @@ typedef __attribute__((ext_vector_type(4))) unsigned short cv_u16x4;
 // borders) come out 0. Vector reads only where provably in-row; the guarded
 // per-element path touches only valid addresses, so no slack allocation or
 // address clamping is needed.
+// BRANCHLESS by construction: the border path issues 8 unconditional
+// clamped scalar loads and zero-selects afterwards — an if-guarded load
+// per element makes hipcc branch around each load and wait vmcnt(0) per
+// element (measured: 175 vmcnt(0) in the staging, ~7x kernel slowdown).
+// Every address stays inside the image: rows clamp into [0, H) and columns
+// into [0, W).
 template <int ST, int SV>
 __device__ __forceinline__ void vstage8(const cvbf16* __restrict__ x,
                                         int H, int W, int ci, int hv,
@@ -116,12 +122,14 @@ __device__ __forceinline__ void vstage8(const cvbf16* __restrict__ x,
         }
       }
     } else {
+      const int wmax = W - 1;
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         const int c = wv0 + i * ST;
-        cvbf16 v = cvf2b(0.f);
-        if (rok && (unsigned)c < (unsigned)W) v = x[row + c];
-        out[i] = *reinterpret_cast<unsigned short*>(&v);
+        const cvbf16 v = x[row + min(max(c, 0), wmax)];
+        out[i] = (rok && (unsigned)c < (unsigned)W)
+                     ? *reinterpret_cast<const unsigned short*>(&v)
+                     : (unsigned short)0;
       }
     }
   } else {  // SV == 2 (zero-stuffed input), ST == 1 always in this regime
@@ -140,13 +148,14 @@ __device__ __forceinline__ void vstage8(const cvbf16* __restrict__ x,
         out[i] = ((i & 1) == par) ? e : (unsigned short)0;
       }
     } else {
+      const int wmax = W - 1;
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         const int c = wv0 + i;
-        cvbf16 v = cvf2b(0.f);
-        if (rok && !(c & 1) && (unsigned)(c >> 1) < (unsigned)W)
-          v = x[row + (c >> 1)];
-        out[i] = *reinterpret_cast<unsigned short*>(&v);
+        const cvbf16 v = x[row + min(max(c >> 1, 0), wmax)];
+        out[i] = (rok && !(c & 1) && (unsigned)(c >> 1) < (unsigned)W)
+                     ? *reinterpret_cast<const unsigned short*>(&v)
+                     : (unsigned short)0;
       }
     }
   }
@@ -161,8 +170,11 @@ __device__ __forceinline__ void vstage8(const cvbf16* __restrict__ x,
 // VM=1: packed-coordinate tables + VIRTUAL pad/stuff — xpad is the RAW
 // (Ci, vH, vW) tensor and staging zero-masks the pad ring and stuff holes
 // (vpt/vpl = virtual pads, vsv = stuff stride). Kills the pad_stuff kernel
-// and the padded-buffer HBM round trip per conv.
-template <int TM, int VM>
+// and the padded-buffer HBM round trip per conv. VST/VSV are the
+// COMPILE-TIME conv stride / stuff stride for VM=1 (VST=0 = generic
+// per-element path for narrow outputs); runtime `stride`/`vsv` args are
+// used by the VM=0 path and must match VST/VSV when VM=1.
+template <int TM, int VM, int VST, int VSV>
 __global__ __launch_bounds__(256)
 void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // see VM note above
                      const cvbf16* __restrict__ wmat,   // (Co, KP64+AP) 0-pad
@@ -241,38 +253,42 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // see VM note above
       const int kci = ko >> 20;
       const int kdh = (ko >> 10) & 1023;
       const int kdw = ko & 1023;
-      if (stride == 1 || stride == 2) {
+      if (VST != 0) {
         cv_u16x8 sa, sb;
         const int hA = mhA + kdh - vpt, wA = mwA + kdw - vpl;
-        const int hB = mhB + kdh - vpt, wB = mwB + kdw - vpl;
-        if (vsv == 2) {
-          vstage8<1, 2>(x, vH, vW, kci, hA, wA, sa);
-          vstage8<1, 2>(x, vH, vW, kci, hB, wB, sb);
-        } else if (stride == 1) {
-          vstage8<1, 1>(x, vH, vW, kci, hA, wA, sa);
-          vstage8<1, 1>(x, vH, vW, kci, hB, wB, sb);
-        } else {
-          vstage8<2, 1>(x, vH, vW, kci, hA, wA, sa);
-          vstage8<2, 1>(x, vH, vW, kci, hB, wB, sb);
-        }
+        vstage8<VST ? VST : 1, VSV>(x, vH, vW, kci, hA, wA, sa);
+        // the trailing-row (B) run only matters for runs that cross an
+        // output row; the vote makes the skip wave-uniform (most tiles
+        // sit inside one row), halving staging work on the common path
+        if (__any(cross < 8)) {
+          const int hB = mhB + kdh - vpt, wB = mwB + kdw - vpl;
+          vstage8<VST ? VST : 1, VSV>(x, vH, vW, kci, hB, wB, sb);
 #pragma unroll
-        for (int i = 0; i < 8; ++i) st[i] = (i < cross) ? sa[i] : sb[i];
-      } else {  // generic (narrow WO): per-element decode + guarded load
+          for (int i = 0; i < 8; ++i) st[i] = (i < cross) ? sa[i] : sb[i];
+        } else {
+          st = sa;
+        }
+      } else {  // generic (narrow WO / odd stride): per-element, branchless
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
           const int mp = mbase[min(m0 + sm8 + i, M - 1)];
           const int hv = (mp >> 16) + kdh - vpt;
           const int wv = (mp & 0xffff) + kdw - vpl;
-          cvbf16 v = cvf2b(0.f);
-          if (vsv == 2) {
-            if (hv >= 0 && !(hv & 1) && (hv >> 1) < vH && wv >= 0 &&
-                !(wv & 1) && (wv >> 1) < vW)
-              v = x[((long long)kci * vH + (hv >> 1)) * vW + (wv >> 1)];
+          bool ok;
+          long long a;
+          if (VSV == 2) {
+            ok = hv >= 0 && !(hv & 1) && (hv >> 1) < vH && wv >= 0 &&
+                 !(wv & 1) && (wv >> 1) < vW;
+            a = ((long long)kci * vH + min(max(hv >> 1, 0), vH - 1)) * vW +
+                min(max(wv >> 1, 0), vW - 1);
           } else {
-            if ((unsigned)hv < (unsigned)vH && (unsigned)wv < (unsigned)vW)
-              v = x[((long long)kci * vH + hv) * vW + wv];
+            ok = (unsigned)hv < (unsigned)vH && (unsigned)wv < (unsigned)vW;
+            a = ((long long)kci * vH + min(max(hv, 0), vH - 1)) * vW +
+                min(max(wv, 0), vW - 1);
           }
-          st[i] = *reinterpret_cast<unsigned short*>(&v);
+          const cvbf16 v = x[a];
+          st[i] = ok ? *reinterpret_cast<const unsigned short*>(&v)
+                     : (unsigned short)0;
         }
       }
       return;
@@ -517,14 +533,24 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
   // vector path is used only when its 40-element read provably stays
   // inside the tensor, else a per-element guarded path runs (first/last
   // rows of the image, left/right border tiles).
+  // whole-tile interiority is uniform across the workgroup: interior tiles
+  // (the vast majority) take a straight-line vector-staging path identical
+  // in cost to the pre-padded vp==0 path; only border tiles pay the
+  // per-row clamp logic. +16 (not +10) on the column bound keeps the
+  // 16-element vector read inside the row.
+  const bool tile_int =
+      vp > 0 && oy0 >= vp && oy0 + 10 - vp <= Hp && ox0 >= vp &&
+      ox0 + 16 - vp <= Wp;
   auto stage = [&](int c, int buf) {
     char* dst = As8 + buf * TB * 2;
     const long long nelem = (long long)Ci * Hp * Wp;
     for (int ridx = tid; ridx < CIC * 10; ridx += 256) {
       const int ci = ridx / 10, y = ridx % 10;
-      if (vp == 0) {
-        const int yy = min(oy0 + y, Hp - 1);  // bottom edge tiles: clamped
-        const cvbf16* g = x + ((long long)(c * CIC + ci) * Hp + yy) * Wp + ox0;
+      if (vp == 0 || tile_int) {
+        const int gy0 = oy0 - vp;  // >= 0 for tile_int; vp==0: plain base
+        const int yy = min(gy0 + y, Hp - 1);  // bottom edge tiles: clamped
+        const cvbf16* g =
+            x + ((long long)(c * CIC + ci) * Hp + yy) * Wp + (ox0 - vp);
         const u16x8 a = *reinterpret_cast<const u16x8*>(g);
         const u16x8 b = *reinterpret_cast<const u16x8*>(g + 8);
 #pragma unroll
@@ -538,8 +564,9 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
       const int gy = oy0 + y - vp;
       const int cbase = ox0 - vp;
       const bool rowin = gy >= 0 && gy < Hp;
-      const long long off =
-          ((long long)(c * CIC + ci) * Hp + (rowin ? gy : 0)) * Wp + cbase;
+      const long long rowb =
+          ((long long)(c * CIC + ci) * Hp + (rowin ? gy : 0)) * Wp;
+      const long long off = rowb + cbase;
       if (rowin && cbase >= 0 && cbase + 10 <= Wp && off + 40 <= nelem) {
         const cvbf16* g = x + off;
         const u16x8 a = *reinterpret_cast<const u16x8*>(g);
@@ -551,11 +578,14 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
               *reinterpret_cast<cvbf16*>(&v);
         }
       } else {
+        // branchless border path: clamped unconditional loads + select
+        // (an if-guarded load per element serializes on vmcnt(0))
 #pragma unroll
         for (int xi = 0; xi < 10; ++xi) {
-          cvbf16 v = cvf2b(0.f);
           const int col = cbase + xi;
-          if (rowin && col >= 0 && col < Wp) v = x[off + xi];
+          const cvbf16 lv = x[rowb + min(max(col, 0), Wp - 1)];
+          const cvbf16 v =
+              (rowin && (unsigned)col < (unsigned)Wp) ? lv : cvf2b(0.f);
           *reinterpret_cast<cvbf16*>(&dst[aoff(y * XT + xi, ci)]) = v;
         }
       }
@@ -648,7 +678,7 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
 // dW[co][k] += sum over the workgroup's pixel chunk of dy[co][m]*A[m][k].
 // GEMM roles: A' (M'=filter taps K) gathered rows, B' = dy columns.
 // Tile: M'64 (taps) x N'64 (couts), K' = pixels chunked by 32.
-template <int VM>
+template <int VM, int VST, int VSV>
 __global__ __launch_bounds__(256)
 void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
                      const cvbf16* __restrict__ dy,    // (Co, M)
@@ -710,46 +740,48 @@ void conv_wrw_kernel(const cvbf16* __restrict__ xpad,  // (Ci, Hp, Wp)
       if (tap < K && p < p1) {
         const int gm0 = min(p, M - 1), gm7 = min(p + 7, M - 1);
         const int mpA = mbase[gm0], mpB = mbase[gm7];
-        const bool gen = (WO < 8) || (st != 1 && st != 2);
-        if (!gen) {
+        if (VST != 0) {
           const int hA = (mpA >> 16) + kdh - vpt;
           const int wA = (mpA & 0xffff) + kdw - vpl;
-          const int hB = (mpB >> 16) + kdh - vpt;
-          const int wB = (mpB & 0xffff) - 7 * st + kdw - vpl;
-          const int cross = WO - (gm0 % WO);
+          // ow is recoverable from the packed field (= ow*VST): no int div
+          const int cross = WO - ((mpA & 0xffff) >> (VST == 2 ? 1 : 0));
           cv_u16x8 va, vb;
-          if (vsv == 2) {
-            vstage8<1, 2>(x, vH, vW, kci, hA, wA, va);
-            vstage8<1, 2>(x, vH, vW, kci, hB, wB, vb);
-          } else if (st == 1) {
-            vstage8<1, 1>(x, vH, vW, kci, hA, wA, va);
-            vstage8<1, 1>(x, vH, vW, kci, hB, wB, vb);
-          } else {
-            vstage8<2, 1>(x, vH, vW, kci, hA, wA, va);
-            vstage8<2, 1>(x, vH, vW, kci, hB, wB, vb);
-          }
+          vstage8<VST ? VST : 1, VSV>(x, vH, vW, kci, hA, wA, va);
+          if (__any(cross < 8)) {
+            const int hB = (mpB >> 16) + kdh - vpt;
+            const int wB = (mpB & 0xffff) - 7 * VST + kdw - vpl;
+            vstage8<VST ? VST : 1, VSV>(x, vH, vW, kci, hB, wB, vb);
 #pragma unroll
-          for (int i = 0; i < 8; ++i)
-            sa[i] = (p + i < p1) ? ((i < cross) ? va[i] : vb[i])
-                                 : (unsigned short)0;
-        } else {
+            for (int i = 0; i < 8; ++i)
+              sa[i] = (p + i < p1) ? ((i < cross) ? va[i] : vb[i])
+                                   : (unsigned short)0;
+          } else {
+#pragma unroll
+            for (int i = 0; i < 8; ++i)
+              sa[i] = (p + i < p1) ? va[i] : (unsigned short)0;
+          }
+        } else {  // generic: per-element decode, branchless clamped loads
 #pragma unroll
           for (int i = 0; i < 8; ++i) {
-            cvbf16 v = cvf2b(0.f);
-            if (p + i < p1) {
-              const int mp = mbase[min(p + i, M - 1)];
-              const int hv = (mp >> 16) + kdh - vpt;
-              const int wv = (mp & 0xffff) + kdw - vpl;
-              if (vsv == 2) {
-                if (hv >= 0 && !(hv & 1) && (hv >> 1) < vH && wv >= 0 &&
-                    !(wv & 1) && (wv >> 1) < vW)
-                  v = x[((long long)kci * vH + (hv >> 1)) * vW + (wv >> 1)];
-              } else {
-                if ((unsigned)hv < (unsigned)vH && (unsigned)wv < (unsigned)vW)
-                  v = x[((long long)kci * vH + hv) * vW + wv];
-              }
+            const int mp = mbase[min(p + i, M - 1)];
+            const int hv = (mp >> 16) + kdh - vpt;
+            const int wv = (mp & 0xffff) + kdw - vpl;
+            bool ok = p + i < p1;
+            long long a;
+            if (VSV == 2) {
+              ok = ok && hv >= 0 && !(hv & 1) && (hv >> 1) < vH && wv >= 0 &&
+                   !(wv & 1) && (wv >> 1) < vW;
+              a = ((long long)kci * vH + min(max(hv >> 1, 0), vH - 1)) * vW +
+                  min(max(wv >> 1, 0), vW - 1);
+            } else {
+              ok = ok && (unsigned)hv < (unsigned)vH &&
+                   (unsigned)wv < (unsigned)vW;
+              a = ((long long)kci * vH + min(max(hv, 0), vH - 1)) * vW +
+                  min(max(wv, 0), vW - 1);
             }
-            sa[i] = *reinterpret_cast<unsigned short*>(&v);
+            const cvbf16 v = x[a];
+            sa[i] = ok ? *reinterpret_cast<const unsigned short*>(&v)
+                       : (unsigned short)0;
           }
         }
       } else {

@@ -170,6 +170,12 @@ class Trainer:
     def _step_inner(self, x: torch.Tensor, y: Optional[torch.Tensor]):
         """zero-grad + forward + backward + both optimizer steps. This is
         what gets graph-captured; it must stay free of host syncs."""
+        # bump the W-panel cache INSIDE the captured region: each step (and
+        # each graph replay) must rebuild panels from the post-optimizer
+        # weights — a bump outside _step_inner would let capture record a
+        # cache hit and replay stale panels forever
+        from ..ops import conv as _conv
+        _conv.begin_step()
         self.opt_ae.zero_grad(set_to_none=not self.use_cuda_graph
                               or self._fused)
         self.opt_pc.zero_grad(set_to_none=not self.use_cuda_graph
@@ -225,8 +231,6 @@ class Trainer:
 
     def train_step(self, x: torch.Tensor, y: Optional[torch.Tensor]):
         """Returns (loss, bpp) tensors on device (no host sync)."""
-        from ..ops import conv as _conv
-        _conv.begin_step()  # weights changed since the last step's optimizer
         self.sched_ae.set_step(self.global_step)
         self.sched_pc.set_step(self.global_step)
 
