@@ -109,6 +109,51 @@ def _plan_v2(device, M: int, K: int, WO: int, stride: int, dil: int,
     return p
 
 
+def _phase_plans(device, HO: int, WO: int, kh: int, kw: int, Cin: int,
+                 pt: int, pl: int):
+    """Phase decomposition of a (st=1, sv=2, dil=1) virtually-stuffed gather
+    (conv-transpose forward / stride-2 backward-data) into 4 DENSE stride-1
+    sub-convs over the raw input: output pixels of parity (a, b) depend only
+    on taps of parity ((pt+a)&1, (pl+b)&1), so each phase is an ordinary
+    (st=1, sv=1) gather on the halved grids — 1/4 the MFMA work and
+    full-vector staging instead of reconstructing stuff holes."""
+    key = ("phase", device.index, HO, WO, kh, kw, Cin, pt, pl)
+    plans = _PLANS.get(key)
+    if plans is not None:
+        return plans
+    plans = []
+    for a in (0, 1):
+        for b in (0, 1):
+            qa = (pt + a) & 1
+            qb = (pl + b) & 1
+            nrs = len(range(qa, kh, 2))
+            nss = len(range(qb, kw, 2))
+            nh = len(range(a, HO, 2))
+            nw = len(range(b, WO, 2))
+            if nrs == 0 or nss == 0 or nh == 0 or nw == 0:
+                continue
+            m = torch.arange(nh * nw, device=device, dtype=torch.int32)
+            mpack = ((torch.div(m, nw, rounding_mode="floor")) << 16) | (m % nw)
+            kk = torch.arange(Cin * nrs * nss, device=device,
+                              dtype=torch.int32)
+            ci = torch.div(kk, nrs * nss, rounding_mode="floor")
+            rem = kk % (nrs * nss)
+            dr = torch.div(rem, nss, rounding_mode="floor")
+            ds = rem % nss
+            kpack = (ci << 20) | (dr << 10) | ds
+            ktab = ci * (kh * kw) + (qa + 2 * dr) * kw + (qb + 2 * ds)
+            plans.append({
+                "mpack": mpack.contiguous(), "kpack": kpack.contiguous(),
+                "ktab": ktab.contiguous(),
+                "ktab64": ktab.to(torch.int64).contiguous(),
+                "nh": nh, "nw": nw, "Kp": int(Cin * nrs * nss),
+                "ptp": (pt - a - qa) // 2, "plp": (pl - b - qb) // 2,
+                "a": a, "b": b,
+            })
+    _PLANS[key] = plans
+    return plans
+
+
 def _dummy_tables(device):
     """Placeholder table args for the direct-kernel branch (unused there)."""
     key = ("dummy", device.index)
@@ -187,11 +232,25 @@ class _GatherConvFn(torch.autograd.Function):
         if direct:
             dt = _dummy_tables(x.device)
             y = ext_fwd(x, _wmat(w1, direct=True), bias32, dt[0], dt[1], Co,
-                        K, HO, WO, act, 1, 1, pt, 0, 0, 0, 1)
+                        K, HO, WO, act, 1, 1, pt, 0, 0, 0, 1,
+                        None, 0, 0, 1, WO)
+        elif sv == 2 and dil == 1:
+            # conv-transpose: 4 dense stride-1 phase sub-convs (see
+            # _phase_plans) writing strided into one full-grid output
+            pg = _require_ext("panel_gather")
+            y = torch.empty(B, Co, HO, WO, dtype=torch.bfloat16,
+                            device=x.device)
+            for p_ in _phase_plans(x.device, HO, WO, kh, kw, Ci, pt, pl):
+                wp = _wmat_cache(
+                    w1, ("ph", p_["a"], p_["b"], kh, kw),
+                    lambda p_=p_: pg(w1.contiguous(), p_["ktab"]))
+                ext_fwd(x, wp, bias32, p_["mpack"], p_["kpack"], Co,
+                        p_["Kp"], p_["nh"], p_["nw"], act, 1, 0, 0, 1,
+                        p_["ptp"], p_["plp"], 1, y, p_["a"], p_["b"], 2, WO)
         else:
             mpack, kpack = _plan_v2(x.device, HO * WO, K, WO, st, dil, kh, kw)
             y = ext_fwd(x, _wmat(w1), bias32, mpack, kpack, Co, K, HO, WO,
-                        act, st, 0, 0, 1, pt, pl, sv)
+                        act, st, 0, 0, 1, pt, pl, sv, None, 0, 0, 1, WO)
         ctx.save_for_backward(x, w1, y if act else None)
         ctx.meta = (st, sv, dil, kh, kw, pt, pl, HO, WO, bias is not None, act)
         return y
@@ -218,19 +277,49 @@ class _GatherConvFn(torch.autograd.Function):
                     dt = _dummy_tables(x.device)
                     dx = ext_fwd(dy, _wmat_rot(w1, 9, True), None, dt[0],
                                  dt[1], Ci, Co * 9, H, W, 0, 1, 1, ptb,
-                                 0, 0, 0, 1)
+                                 0, 0, 0, 1, None, 0, 0, 1, W)
+                elif st == 2 and dil == 1:
+                    # stride-2 backward-data: phase sub-convs over raw dy
+                    # with per-phase column-gathers of the rotated panel
+                    pg = _require_ext("panel_gather")
+                    rot = _wmat_rot(w1, kh * kw)
+                    dx = torch.empty(B, Ci, H, W, dtype=torch.bfloat16,
+                                     device=x.device)
+                    for p_ in _phase_plans(dy.device, H, W, kh, kw, Co,
+                                           ptb, plb):
+                        wp = _wmat_cache(
+                            w1, ("phr", p_["a"], p_["b"], kh, kw),
+                            lambda p_=p_: pg(rot, p_["ktab"]))
+                        ext_fwd(dy, wp, None, p_["mpack"], p_["kpack"], Ci,
+                                p_["Kp"], p_["nh"], p_["nw"], 0, 1, 0, 0, 1,
+                                p_["ptp"], p_["plp"], 1, dx, p_["a"],
+                                p_["b"], 2, W)
                 else:
                     mb, kb = _plan_v2(dy.device, H * W, Co * kh * kw, W, sv,
                                       dil, kh, kw)
                     dx = ext_fwd(dy, _wmat_rot(w1, kh * kw), None, mb, kb,
                                  Ci, Co * kh * kw, H, W, 0, sv, 0, 0,
-                                 1, ptb, plb, st)
+                                 1, ptb, plb, st, None, 0, 0, 1, W)
 
         dw1 = None
         if ctx.needs_input_grad[1]:
-            mpack, kpack = _plan_v2(x.device, HO * WO, K, WO, st, dil, kh, kw)
-            dw1 = ext_wrw(x, dy, mpack, kpack, Co, K, WO, st == 1,
-                          1, st, pt, pl, sv).to(w1.dtype)
+            if sv == 2 and dil == 1:
+                # conv-transpose weight grad by phases: each phase's compact
+                # dW lands in its own tap columns (disjoint across phases)
+                dw32 = torch.zeros(Co, K, dtype=torch.float32,
+                                   device=x.device)
+                for p_ in _phase_plans(x.device, HO, WO, kh, kw, Ci, pt, pl):
+                    dy_p = dy[:, :, p_["a"]::2, p_["b"]::2].contiguous()
+                    dwp = ext_wrw(x, dy_p, p_["mpack"], p_["kpack"], Co,
+                                  p_["Kp"], p_["nw"], True, 1, 1,
+                                  p_["ptp"], p_["plp"], 1)
+                    dw32.index_copy_(1, p_["ktab64"], dwp)
+                dw1 = dw32.to(w1.dtype)
+            else:
+                mpack, kpack = _plan_v2(x.device, HO * WO, K, WO, st, dil,
+                                        kh, kw)
+                dw1 = ext_wrw(x, dy, mpack, kpack, Co, K, WO, st == 1,
+                              1, st, pt, pl, sv).to(w1.dtype)
 
         dbias = (dy.sum(dim=(0, 2, 3), dtype=torch.float32)
                  if has_bias else None)
@@ -256,7 +345,7 @@ class _GatherConvFP8Fn(torch.autograd.Function):
         mbase, koff = _plan(x.device, Ci, Hp, Wp, kh, kw, stride, dil, HO, WO)
         bias32 = bias.float().contiguous() if bias is not None else None
         y = ext_fwd(xbuf, _wmat(w1, fp8=True), bias32, mbase, koff, Co, K,
-                    HO, WO, act, stride, 0, 0, 0, 0, 0, 1)
+                    HO, WO, act, stride, 0, 0, 0, 0, 0, 1, None, 0, 0, 1, WO)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (stride, dil, kh, kw, HO, WO, bias is not None, act,
                     pads, stuff, x.shape, x.dtype)
@@ -293,7 +382,7 @@ class _GatherConvFP8Fn(torch.autograd.Function):
                                  kh, kw, 1, dil, Hp, Wp)
                 dxbuf = ext_fwd(dybuf, _wmat(wrot, fp8=True), None, mb2, ko2,
                                 Ci, Co * kh * kw, Hp, Wp, 0, 1, 0, 0,
-                                0, 0, 0, 1)
+                                0, 0, 0, 1, None, 0, 0, 1, Wp)
                 _, _, H, W = xshape
                 dx = dxbuf[:, :, pt:pt + (H - 1) * stuff + 1:stuff,
                            pl:pl + (W - 1) * stuff + 1:stuff].to(xdtype)
@@ -422,7 +511,8 @@ class _GatherConv3dFn(torch.autograd.Function):
         bias32 = bias.float().contiguous() if bias is not None else None
         # kernel sees a 2D problem: M = Do*Ho*Wo pixels, "WO" = Wo rows
         y = ext_fwd(xbuf.view(B, Ci, Dp * Hp, Wp), _wmat(w1), bias32, mbase,
-                    koff, Co, K, Do * Ho, Wo, act, 1, 0, 0, 0, 0, 0, 1)
+                    koff, Co, K, Do * Ho, Wo, act, 1, 0, 0, 0, 0, 0, 1,
+                    None, 0, 0, 1, Wo)
         ctx.save_for_backward(xbuf, w1, y if act else None)
         ctx.meta = (kd, kh, kw, Do, Ho, Wo, bias is not None, act)
         return y.view(B, Co, Do, Ho, Wo)
@@ -448,7 +538,8 @@ class _GatherConv3dFn(torch.autograd.Function):
                 dxbuf = ext_fwd(dybuf.view(B, Co, -1, dybuf.shape[4]),
                                 _wmat_rot(w1, kd * kh * kw), None,
                                 mb2, ko2, Ci, Co * kd * kh * kw, Dp * Hp, Wp,
-                                0, 1, 0, 0, 0, 0, 0, 1).view(B, Ci, Dp, Hp, Wp)
+                                0, 1, 0, 0, 0, 0, 0, 1,
+                                None, 0, 0, 1, Wp).view(B, Ci, Dp, Hp, Wp)
 
         dw1 = None
         if ctx.needs_input_grad[1]:

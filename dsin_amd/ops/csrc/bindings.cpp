@@ -28,7 +28,9 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        torch::Tensor koff, int64_t N, int64_t K, int64_t HO,
                        int64_t WO, int64_t act, int64_t stride, int64_t direct,
                        int64_t vpad, int64_t vm, int64_t vpt, int64_t vpl,
-                       int64_t vsv);
+                       int64_t vsv, c10::optional<torch::Tensor> out_opt,
+                       int64_t oh0, int64_t ow0, int64_t ostep, int64_t wof);
+torch::Tensor panel_gather(torch::Tensor src, torch::Tensor ktab);
 torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t WO, bool mcontig, int64_t vm,
@@ -67,6 +69,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_wrw", &dsin::conv_wrw, "implicit-GEMM conv weight gradient");
   m.def("act_bwd", &dsin::act_bwd, "fused activation gradient (bf16 out)");
   m.def("wmat_make", &dsin::wmat_make, "padded/rotated bf16 conv W panel");
+  m.def("panel_gather", &dsin::panel_gather,
+        "column-gather of a W panel (phase-decomposed convs)");
   m.def("adam_step", &dsin::adam_step, "fused flat-buffer Adam step");
   m.def("pad_stuff", &dsin::pad_stuff, "fused pad/zero-stuff/cast to bf16");
   m.def("bn_fwd", &dsin::bn_fwd, "fused batch-norm(+act) forward");

@@ -44,7 +44,9 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        torch::Tensor mbase, torch::Tensor koff, int64_t N,
                        int64_t K, int64_t HO, int64_t WO, int64_t act,
                        int64_t stride, int64_t direct, int64_t vpad,
-                       int64_t vm, int64_t vpt, int64_t vpl, int64_t vsv) {
+                       int64_t vm, int64_t vpt, int64_t vpl, int64_t vsv,
+                       c10::optional<torch::Tensor> out_opt, int64_t oh0,
+                       int64_t ow0, int64_t ostep, int64_t wof) {
   CHECK_CUDA_CONTIG(xbuf);
   CHECK_CUDA_CONTIG(wmat);
   const bool fp8 = xbuf.scalar_type() == torch::kByte;  // raw e4m3 bytes
@@ -85,8 +87,24 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   const int64_t KP = (K + 63) & ~63;  // 64-chunk padded; wmat zero-padded
   TORCH_CHECK(wmat.size(1) == KP + CONV_AP, "wmat row stride mismatch");
   if (WO < 8) stride = 0;  // scalar staging path for very narrow outputs
-  auto out = torch::empty({B, N, HO, WO},
-                          xbuf.options().dtype(torch::kBFloat16));
+  // phase-strided output: write into a caller-provided full-grid tensor
+  torch::Tensor out;
+  long long o_chan = M, o_img = (long long)N * M;
+  int WOf = (int)WO;
+  if (out_opt.has_value()) {
+    out = out_opt.value();
+    CHECK_CUDA_CONTIG(out);
+    TORCH_CHECK(out.scalar_type() == torch::kBFloat16 && out.dim() == 4 &&
+                    out.size(0) == B && out.size(1) == N,
+                "phase out tensor mismatch");
+    o_chan = (long long)out.size(2) * out.size(3);
+    o_img = (long long)N * o_chan;
+    WOf = (int)out.size(3);
+  } else {
+    TORCH_CHECK(ostep == 1, "strided output needs an out tensor");
+    out = torch::empty({B, N, HO, WO},
+                       xbuf.options().dtype(torch::kBFloat16));
+  }
   const float* bptr = nullptr;
   if (bias.has_value()) {
     CHECK_CUDA_CONTIG(bias.value());
@@ -94,6 +112,7 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   }
   dim3 grid((M + CONV_TM - 1) / CONV_TM, (N + CONV_TN - 1) / CONV_TN, B);
   if (fp8) {
+    TORCH_CHECK(!out_opt.has_value(), "fp8 path has no phase output");
     size_t lds = (size_t)2 * CONV8_TM * (64 + CONV8_AP);
     hipLaunchKernelGGL(conv_fwd_fp8_kernel, grid, dim3(256), lds,
                        at::hip::getCurrentHIPStreamMasqueradingAsCUDA(),
@@ -111,7 +130,8 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   using FwdKern = void (*)(const cvbf16*, const cvbf16*, const float*,
                            cvbf16*, const int*, const int*, int, int, int,
                            int, long long, long long, int, int, int, int,
-                           int, int, int, int);
+                           int, int, int, int, int, int, int, int,
+                           long long);
   auto pick_fwd = [&](bool tm32) -> FwdKern {
     if (!vm) return tm32 ? conv_fwd_kernel<32, 0, 0, 1>
                          : conv_fwd_kernel<64, 0, 0, 1>;
@@ -141,8 +161,9 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                        (const cvbf16*)wmat.data_ptr(), bptr,
                        (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
                        koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
-                       xbuf.stride(0), (long long)N * M, (int)act, (int)WO,
-                       (int)stride, vH, vW, (int)vpt, (int)vpl, (int)vsv);
+                       xbuf.stride(0), o_img, (int)act, (int)WO,
+                       (int)stride, vH, vW, (int)vpt, (int)vpl, (int)vsv,
+                       (int)oh0, (int)ow0, (int)ostep, WOf, o_chan);
     return out;
   }
   size_t lds = (size_t)4 * CONV_TM * (64 + CONV_AP) * 2;  // 4-buffer pipeline
@@ -152,8 +173,9 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
                      (const cvbf16*)wmat.data_ptr(), bptr,
                      (cvbf16*)out.data_ptr(), mbase.data_ptr<int>(),
                      koff.data_ptr<int>(), (int)M, (int)N, (int)K, (int)KP,
-                     xbuf.stride(0), (long long)N * M, (int)act, (int)WO,
-                     (int)stride, vH, vW, (int)vpt, (int)vpl, (int)vsv);
+                     xbuf.stride(0), o_img, (int)act, (int)WO,
+                     (int)stride, vH, vW, (int)vpt, (int)vpl, (int)vsv,
+                     (int)oh0, (int)ow0, (int)ostep, WOf, o_chan);
   return out;
 }
 
@@ -223,6 +245,31 @@ torch::Tensor conv_wrw(torch::Tensor xbuf, torch::Tensor dy,
   }
   if (B * pix_chunks == 1) return dwp.view({N, K});
   return dwp.sum(0);
+}
+
+torch::Tensor panel_gather(torch::Tensor src, torch::Tensor ktab) {
+  CHECK_CUDA_CONTIG(src);
+  CHECK_CUDA_CONTIG(ktab);
+  const int64_t rows = src.size(0);
+  const int64_t Kl = ktab.numel();
+  const int64_t KPA = ((Kl + 63) & ~63) + CONV_AP;
+  auto out = torch::empty({rows, KPA}, src.options().dtype(torch::kBFloat16));
+  const long long total = rows * KPA;
+  const int grid = (int)std::min<long long>((total + 255) / 256, 4096);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (src.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL((panel_gather_kernel<float>), dim3(grid), dim3(256), 0,
+                       stream, src.data_ptr<float>(), ktab.data_ptr<int>(),
+                       (cvbf16*)out.data_ptr(), (int)rows, (int)Kl, (int)KPA,
+                       (int)src.size(1));
+  } else {
+    TORCH_CHECK(src.scalar_type() == torch::kBFloat16, "panel src dtype");
+    hipLaunchKernelGGL((panel_gather_kernel<cvbf16>), dim3(grid), dim3(256),
+                       0, stream, (const cvbf16*)src.data_ptr(),
+                       ktab.data_ptr<int>(), (cvbf16*)out.data_ptr(),
+                       (int)rows, (int)Kl, (int)KPA, (int)src.size(1));
+  }
+  return out;
 }
 
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor y, int64_t act) {

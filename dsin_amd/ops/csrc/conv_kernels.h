@@ -186,7 +186,12 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // see VM note above
                      long long x_img_stride,            // Ci*Hp*Wp
                      long long o_img_stride,            // Co*M
                      int act, int WO,
-                     int stride, int vH, int vW, int vpt, int vpl, int vsv) {
+                     int stride, int vH, int vW, int vpt, int vpl, int vsv,
+                     int oh0, int ow0, int ostep, int WOf,
+                     long long o_chan) {
+  // oh0/ow0/ostep/WOf/o_chan: phase-strided output placement — output pixel
+  // m lands at (oh0 + ostep*(m/WO), ow0 + ostep*(m%WO)) of a WOf-wide image
+  // whose channel stride is o_chan. Plain calls pass (0, 0, 1, WO, M).
   // Branchless pipeline: K chunks of 64 (two MFMA k-steps per barrier),
   // double-buffered LDS A-tile with XOR-swizzled addressing. Staging is two
   // overlapped 16B loads per 8-pixel run with a per-element crossing select
@@ -451,7 +456,8 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // see VM note above
 
   // ---- epilogue: D[row=pixel][col=cout]; row = mi*16 + kgrp*4 + reg ----
   const float bv = (bias != nullptr && ncol < N) ? bias[ncol] : 0.f;
-  cvbf16* o = out + img * o_img_stride + (long long)(ncol < N ? ncol : 0) * M;
+  cvbf16* o = out + img * o_img_stride +
+              (long long)(ncol < N ? ncol : 0) * o_chan;
   if (ncol < N) {
 #pragma unroll
     for (int mi = 0; mi < MI; ++mi) {
@@ -462,7 +468,13 @@ void conv_fwd_kernel(const cvbf16* __restrict__ xpad,   // see VM note above
           float v = acc[mi][reg] + bv;
           if (act == 1) v = fmaxf(v, 0.f);
           else if (act == 2) v = fmaxf(v, 0.2f * v);
-          o[m] = cvf2b(v);
+          if (ostep == 1) {
+            o[m] = cvf2b(v);
+          } else {
+            const int oy = oh0 + ostep * (m / WO);
+            const int ox = ow0 + ostep * (m % WO);
+            o[(long long)oy * WOf + ox] = cvf2b(v);
+          }
         }
       }
     }
@@ -907,6 +919,28 @@ __global__ void wmat_make_kernel(const T* __restrict__ w1,
         v = (float)w1[(long long)n * kin + k];
       }
     }
+    out[i] = cvf2b(v);
+  }
+}
+
+// Column-gather of a weight panel: out[n][kp] = src[n][ktab[kp]] (0 beyond
+// Klocal). Builds the per-phase W panels of the phase-decomposed
+// conv-transpose / stride-2 backward path from either the raw fp32 weight
+// (src_stride = kin) or a cached bf16 rotated panel (src_stride = KPA).
+template <typename T>
+__global__ void panel_gather_kernel(const T* __restrict__ src,
+                                    const int* __restrict__ ktab,
+                                    cvbf16* __restrict__ out,
+                                    int rows, int Klocal, int KPA,
+                                    int src_stride) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long total = (long long)rows * KPA;
+  const long long gstride = (long long)gridDim.x * blockDim.x;
+  for (; i < total; i += gstride) {
+    const int n = (int)(i / KPA);
+    const int k = (int)(i % KPA);
+    float v = 0.f;
+    if (k < Klocal) v = (float)src[(long long)n * src_stride + ktab[k]];
     out[i] = cvf2b(v);
   }
 }
