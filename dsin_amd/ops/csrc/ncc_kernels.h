@@ -316,6 +316,229 @@ void ncc_main_kernel(const ncbf16* __restrict__ tx, const ncbf16* __restrict__ t
   }
 }
 
+// ------------------------------------------------------- stage 4, variant 2
+// Requires pw % 8 == 0 (the default (20,24) and every bench fallback patch).
+// The v1 kernel keeps the whole (32 x K) A-tile in LDS (93 KB at K=1440 ->
+// ONE workgroup per CU, 1 wave/SIMD) and gathers each B fragment with 8
+// scalar LDS reads + packing (issue-bound, 105 cyc/MFMA measured). Here:
+//   * A is staged in K-SLICES of NCC_SL taps (33 KB -> 3-4 workgroups/CU,
+//     the load latency finally has TLP to hide under);
+//   * each 8-element k-group of a row lies in ONE patch row, so a B
+//     fragment is a single unaligned global u16x8 read of the L1/L2-hot
+//     transformed side image (no y-window staging, no offset table, no
+//     packing VALU);
+//   * the k-loop is OUTER over 4-row output groups, so one LDS A fragment
+//     feeds 4 MFMAs (4 independent accumulator chains).
+constexpr int NCC_SL = 384;  // taps per A slice (multiple of 16; 2x25KB buffers -> 3 blocks/CU)
+constexpr int NCC_TJ2 = 64;  // v2 col tile: 2 j-waves x 32 — the y window
+                             // (3 x 27 x 87 bf16 = 14 KB/block) then fits
+                             // L1 with two co-resident blocks; the other 2
+                             // waves take the other 4-row output group
+
+__global__ __launch_bounds__(256)
+void ncc_main_v2_kernel(const ncbf16* __restrict__ tx,
+                        const ncbf16* __restrict__ ty,
+                        const unsigned int* __restrict__ aoffs,
+                        const float* __restrict__ psum,
+                        const float* __restrict__ psum2,
+                        const float* __restrict__ sy,
+                        const float* __restrict__ sy2,
+                        unsigned long long* __restrict__ best,  // (P,)
+                        int H, int W, int ph, int pw, int gw, int P,
+                        int Hc, int Wc, int use_mask) {
+  const int K = 3 * ph * pw;
+  const int KP = (K + 15) & ~15;
+  const int ASTRIDE = NCC_SL + NCC_APAD;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  ncbf16* As = reinterpret_cast<ncbf16*>(smem);      // 2 x [TP][ASTRIDE]
+  float* Pstat = reinterpret_cast<float*>(As + 2 * NCC_TP * ASTRIDE);  // [5][TP]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int j0 = blockIdx.x * NCC_TJ2;
+  const int i0 = blockIdx.y * NCC_TI;
+  const int p0 = blockIdx.z * NCC_TP;
+
+  const float fK = (float)K;
+  const float invK = 1.f / fK;
+  const float ish2 = 4.0f / ((float)H * (float)H);
+  const float isw2 = 4.0f / ((float)W * (float)W);
+
+  if (tid < NCC_TP) {
+    const int p = min(p0 + tid, P - 1);
+    const float sxv = psum[p];
+    const float xm = sxv * invK;
+    Pstat[tid] = sxv;
+    Pstat[NCC_TP + tid] = xm;
+    Pstat[2 * NCC_TP + tid] = psum2[p] - 2.f * xm * sxv + fK * xm * xm;
+    Pstat[3 * NCC_TP + tid] = ((float)(p / gw) + 0.5f) * (float)ph;
+    Pstat[4 * NCC_TP + tid] = ((float)(p % gw) + 0.5f) * (float)pw;
+  }
+
+  // slice staging: 8 threads per patch row, each stages 64 taps (8 runs of
+  // 8 contiguous image elements via aoffs); zero-fills beyond K / beyond P
+  const int s_pi = tid & 31;
+  const int s_k0 = (tid >> 5) * 64;
+  const int sp = p0 + s_pi;
+  const bool s_pok = sp < P;
+  const long long s_pbase =
+      s_pok ? (long long)((sp / gw) * ph) * W + (long long)((sp % gw) * pw)
+            : 0;
+  typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+  auto stage_slice = [&](int s0, int buf) {
+    ncbf16* dst = As + buf * NCC_TP * ASTRIDE;
+#pragma unroll
+    for (int r8 = 0; r8 < 8; ++r8) {
+      const int kl = s_k0 + r8 * 8;
+      const int kg = s0 + kl;
+      u16x8 v = {};
+      if (s_pok && kg < K)
+        v = *reinterpret_cast<const u16x8*>(&tx[s_pbase + aoffs[kg]]);
+      *reinterpret_cast<u16x8*>(&dst[s_pi * ASTRIDE + kl]) = v;
+    }
+  };
+
+  const int colL = lane & 31;
+  const int kgrp = lane >> 5;
+  const int jj = (wid & 1) * 32 + colL;   // 2 j-waves cover TJ2 columns
+  const int g0 = (wid >> 1) * 4;          // 2 row-group waves cover TI rows
+  const int j = j0 + jj;
+  const bool jvalid = j < Wc;
+
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  unsigned long long bestk[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) bestk[r] = 0ull;
+
+  const int nslices = (KP + NCC_SL - 1) / NCC_SL;
+
+  // per-row epilogue (identical math to the v1 kernel)
+  auto epilogue_row = [&](const f32x16& acc, int iw) {
+    const long long sidx = (long long)iw * Wc + (jvalid ? j : 0);
+    const float syv = sy[sidx];
+    const float sy2v = sy2[sidx];
+    const float ym = syv * invK;
+    const float deny = sy2v - 2.f * ym * syv + fK * ym * ym;
+    const float di = (float)(iw + ph / 2 - 1);
+    const unsigned int idx = (unsigned int)(iw * Wc + j);
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int prow = (reg & 3) + 8 * (reg >> 2) + 4 * kgrp;
+      const float av = acc[reg];
+      const float sxv = Pstat[prow];
+      const float xm = Pstat[NCC_TP + prow];
+      const float denx = Pstat[2 * NCC_TP + prow];
+      const float num = av - ym * sxv - xm * syv + fK * xm * ym;
+      float val = num * __builtin_amdgcn_rsqf(fmaxf(denx * deny, NCC_EPS));
+      if (use_mask) {
+        const float dd = di - Pstat[3 * NCC_TP + prow];
+        const float dj = (float)(j + pw / 2 - 1) - Pstat[4 * NCC_TP + prow];
+        val *= __expf(-FOURLN2 * (dd * dd * ish2 + dj * dj * isw2));
+      }
+      unsigned long long key = ((unsigned long long)nfloat_flip(val) << 32) |
+                               (unsigned long long)(~idx);
+      key = (jvalid && (p0 + prow) < P) ? key : 0ull;
+      if (key > bestk[reg]) bestk[reg] = key;
+    }
+  };
+
+  // this wave's 4-row output group (the A slices are shared by all waves)
+  const int iig = i0 + g0;
+  // clamped row offsets (rows beyond Hc read row Hc-1; discarded below)
+  long long rofs[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    rofs[r] = (long long)min(iig + r, Hc - 1) * W + (j0 + jj);
+  f32x16 a0 = {}, a1 = {}, a2 = {}, a3 = {};
+  // double-buffered slices: slice s+1 stages (scattered global reads + LDS
+  // writes into the other buffer) while the MFMA loop consumes slice s; one
+  // barrier per slice
+  stage_slice(0, 0);
+  __syncthreads();
+#pragma unroll 1
+  for (int s = 0; s < nslices; ++s) {
+    if (s + 1 < nslices) stage_slice((s + 1) * NCC_SL, (s + 1) & 1);
+    ncbf16* As_cur = As + (s & 1) * NCC_TP * ASTRIDE;
+    // incremental (c, a, b) decode of this half-wave's k-group
+    int dyc, dya, dyb;
+    {
+      int k0g = s * NCC_SL + kgrp * 8;
+      dyc = k0g / (ph * pw);
+      int rem = k0g % (ph * pw);
+      dya = rem / pw;
+      dyb = rem % pw;
+    }
+    const int nkt = min(NCC_SL, KP - s * NCC_SL) / 16;
+    auto advance = [&]() {
+      dyb += 16;
+      if (dyb >= pw) {
+        dyb -= pw;
+        if (++dya >= ph) { dya = 0; ++dyc; }
+        if (dyb >= pw) {  // pw == 8: one 16-step crosses two rows
+          dyb -= pw;
+          if (++dya >= ph) { dya = 0; ++dyc; }
+        }
+      }
+      if (dyc > 2) { dyc = 2; dya = 0; }  // KP padding: A zeros cancel
+    };
+    // two register SETS ping-pong (no cross-iteration copies): the next
+    // chunk's five loads issue before the current chunk's MFMAs
+    u16x8 v0a, v1a, v2a, v3a, v0b, v1b, v2b, v3b;
+    bf16x8 afa, afb;
+    auto load_set = [&](int kt, u16x8& v0, u16x8& v1, u16x8& v2, u16x8& v3,
+                        bf16x8& af) {
+      const long long offk = ((long long)(dyc * H) + dya) * W + dyb;
+      af = *reinterpret_cast<const bf16x8*>(
+          &As_cur[colL * ASTRIDE + kt * 16 + kgrp * 8]);
+      v0 = *reinterpret_cast<const u16x8*>(&ty[offk + rofs[0]]);
+      v1 = *reinterpret_cast<const u16x8*>(&ty[offk + rofs[1]]);
+      v2 = *reinterpret_cast<const u16x8*>(&ty[offk + rofs[2]]);
+      v3 = *reinterpret_cast<const u16x8*>(&ty[offk + rofs[3]]);
+      advance();
+    };
+    auto mfma_set = [&](const u16x8& v0, const u16x8& v1, const u16x8& v2,
+                        const u16x8& v3, const bf16x8& af) {
+      a0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          af, *reinterpret_cast<const bf16x8*>(&v0), a0, 0, 0, 0);
+      a1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          af, *reinterpret_cast<const bf16x8*>(&v1), a1, 0, 0, 0);
+      a2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          af, *reinterpret_cast<const bf16x8*>(&v2), a2, 0, 0, 0);
+      a3 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          af, *reinterpret_cast<const bf16x8*>(&v3), a3, 0, 0, 0);
+    };
+    load_set(0, v0a, v1a, v2a, v3a, afa);
+    int kt = 0;
+    for (; kt + 2 <= nkt; kt += 2) {
+      if (kt + 1 < nkt) load_set(kt + 1, v0b, v1b, v2b, v3b, afb);
+      mfma_set(v0a, v1a, v2a, v3a, afa);
+      if (kt + 2 < nkt) load_set(kt + 2, v0a, v1a, v2a, v3a, afa);
+      mfma_set(v0b, v1b, v2b, v3b, afb);
+    }
+    if (kt < nkt) mfma_set(v0a, v1a, v2a, v3a, afa);  // odd tail
+    __syncthreads();  // next slice's buffer fully written AND consumed
+  }
+  if (iig + 0 < Hc) epilogue_row(a0, iig + 0);
+  if (iig + 1 < Hc) epilogue_row(a1, iig + 1);
+  if (iig + 2 < Hc) epilogue_row(a2, iig + 2);
+  if (iig + 3 < Hc) epilogue_row(a3, iig + 3);
+
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    unsigned long long k = bestk[reg];
+#pragma unroll
+    for (int off = 16; off > 0; off >>= 1) {
+      unsigned long long other =
+          (unsigned long long)__shfl_xor((long long)k, off, 32);
+      if (other > k) k = other;
+    }
+    const int prow = p0 + (reg & 3) + 8 * (reg >> 2) + 4 * kgrp;
+    if (colL == 0 && prow < P && k != 0ull) atomicMax(&best[prow], k);
+  }
+}
+
 // ---------------------------------------------------------------- stage 5
 
 __global__ void scatter_kernel(const unsigned long long* __restrict__ best,

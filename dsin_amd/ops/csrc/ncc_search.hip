@@ -39,11 +39,15 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ncc_search(
   const int K = 3 * ph * pw;
   const int KP = (K + 15) & ~15;
   const int YR = NCC_TI + ph - 1, YC = NCC_TJ + pw - 1, YCP = (YC + 8) & ~7;
-  const size_t lds = (size_t)NCC_TP * (KP + NCC_APAD) * 2 +
-                     (size_t)3 * YR * YCP * 2 + (size_t)((KP + 7) & ~7) * 2 +
-                     (size_t)5 * NCC_TP * 4;
+  // v2 (pw % 8 == 0, the default geometry): k-sliced LDS A-tile + direct
+  // global B loads — see ncc_main_v2_kernel
+  const bool v2 = (pw % 8) == 0;
+  const size_t lds =
+      v2 ? (size_t)2 * NCC_TP * (NCC_SL + NCC_APAD) * 2 + (size_t)5 * NCC_TP * 4
+         : (size_t)NCC_TP * (KP + NCC_APAD) * 2 + (size_t)3 * YR * YCP * 2 +
+               (size_t)((KP + 7) & ~7) * 2 + (size_t)5 * NCC_TP * 4;
   TORCH_CHECK(lds <= 160 * 1024, "patch size too large for LDS tiling: ", lds);
-  TORCH_CHECK(3 * YR * YCP < 65536, "y-window exceeds u16 offset range");
+  TORCH_CHECK(v2 || 3 * YR * YCP < 65536, "y-window exceeds u16 offset range");
 
   auto optsF = x_dec.options();
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -56,7 +60,10 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ncc_search(
                      K);
 
   auto t_x = torch::empty({3, H, W}, optsF.dtype(torch::kBFloat16));
-  auto t_y = torch::empty({3, H, W}, optsF.dtype(torch::kBFloat16));
+  // +192 elements of tail slack: the v2 kernel's invalid-column vector
+  // loads may overshoot the last row (values discarded)
+  auto t_y = torch::empty({(int64_t)3 * H * W + 192},
+                          optsF.dtype(torch::kBFloat16));
   int64_t hw = (int64_t)H * W;
   hipLaunchKernelGGL(transform_kernel, grid1d(hw, 256), dim3(256), 0, stream,
                      x_dec.data_ptr<float>(), (ncbf16*)t_x.data_ptr(), hw);
@@ -83,17 +90,29 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ncc_search(
                      sy.data_ptr<float>(), sy2.data_ptr<float>(), Hc, Wc, ph);
 
   auto best = torch::zeros({P}, optsF.dtype(torch::kInt64));  // u64 keys
-  dim3 grid((Wc + NCC_TJ - 1) / NCC_TJ, (Hc + NCC_TI - 1) / NCC_TI,
+  const int tj = v2 ? NCC_TJ2 : NCC_TJ;
+  dim3 grid((Wc + tj - 1) / tj, (Hc + NCC_TI - 1) / NCC_TI,
             (P + NCC_TP - 1) / NCC_TP);
-  hipLaunchKernelGGL(ncc_main_kernel, grid, dim3(256), lds, stream,
-                     (const ncbf16*)t_x.data_ptr(),
-                     (const ncbf16*)t_y.data_ptr(),
-                     (const unsigned int*)aoffs.data_ptr(),
-                     (const unsigned short*)koffs.data_ptr(),
-                     psum.data_ptr<float>(), psum2.data_ptr<float>(),
-                     sy.data_ptr<float>(), sy2.data_ptr<float>(),
-                     (unsigned long long*)best.data_ptr(), H, W, ph, pw, gw, P,
-                     Hc, Wc, use_mask ? 1 : 0);
+  if (v2) {
+    hipLaunchKernelGGL(ncc_main_v2_kernel, grid, dim3(256), lds, stream,
+                       (const ncbf16*)t_x.data_ptr(),
+                       (const ncbf16*)t_y.data_ptr(),
+                       (const unsigned int*)aoffs.data_ptr(),
+                       psum.data_ptr<float>(), psum2.data_ptr<float>(),
+                       sy.data_ptr<float>(), sy2.data_ptr<float>(),
+                       (unsigned long long*)best.data_ptr(), H, W, ph, pw,
+                       gw, P, Hc, Wc, use_mask ? 1 : 0);
+  } else {
+    hipLaunchKernelGGL(ncc_main_kernel, grid, dim3(256), lds, stream,
+                       (const ncbf16*)t_x.data_ptr(),
+                       (const ncbf16*)t_y.data_ptr(),
+                       (const unsigned int*)aoffs.data_ptr(),
+                       (const unsigned short*)koffs.data_ptr(),
+                       psum.data_ptr<float>(), psum2.data_ptr<float>(),
+                       sy.data_ptr<float>(), sy2.data_ptr<float>(),
+                       (unsigned long long*)best.data_ptr(), H, W, ph, pw,
+                       gw, P, Hc, Wc, use_mask ? 1 : 0);
+  }
 
   auto y_syn = torch::empty_like(y_orig);
   auto rows = torch::empty({P}, optsF.dtype(torch::kInt64));
