@@ -45,6 +45,9 @@ def main(argv=None):
                     default=os.path.join(cur, "run_configs", "pc_run_configs"))
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--bf16", action="store_true", help="bf16 autocast compute")
+    ap.add_argument("--graphs", action="store_true",
+                    help="hipGraph whole-step capture (same engine path as "
+                         "bench.py; validation runs eager between replays)")
     ap.add_argument("--metrics", type=str, default=None, help="JSONL metrics path")
     ap.add_argument("--checkpoint-every", type=int, default=0,
                     help="periodic crash-recovery checkpoint interval "
@@ -72,7 +75,7 @@ def main(argv=None):
     data = make_dataset(ae_config, data_dir, seed=rank())
     num_train = len(data.train_pairs)
     trainer = Trainer(model, ae_config, pc_config, num_train, device=device,
-                     autocast_bf16=args.bf16)
+                     autocast_bf16=args.bf16, use_cuda_graph=args.graphs)
     metrics = MetricsLogger(args.metrics if rank() == 0 else None)
 
     model_name = "NA"
