@@ -199,9 +199,12 @@ __global__ void bn_bwd_part_kernel(const bnbf16* __restrict__ dy,
     }
   }
   block_reduce2(a, b2);
+  // plain per-slice stores (deterministic: the host reduces the (S, C)
+  // partials with one ordered at::sum — fp32 atomicAdd order varies
+  // between runs and was the one nondeterminism in the BN backward)
   if (threadIdx.x == 0) {
-    atomicAdd(&s1[c], a);
-    atomicAdd(&s2[c], b2);
+    s1[(long long)sidx * C + c] = a;
+    s2[(long long)sidx * C + c] = b2;
   }
 }
 
@@ -314,16 +317,18 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
   const int B = (int)y.size(0), C = (int)y.size(1);
   const long long HW = (long long)y.size(2) * y.size(3);
   auto optsF = y.options().dtype(torch::kFloat32);
-  auto sbuf = torch::zeros({2, C}, optsF);  // one fill for both sums
-  auto s1 = sbuf[0];
-  auto s2 = sbuf[1];
   auto stream = at::cuda::getCurrentCUDAStream();
   const int S = _spatial_chunks(HW);
+  // (2, S, C) plain-store partials -> one ordered sum (deterministic)
+  auto parts = torch::empty({2, S, C}, optsF);
   hipLaunchKernelGGL(bn_bwd_part_kernel, dim3(S, C), dim3(256), 0, stream,
                      (const bnbf16*)dy.data_ptr(), (const bnbf16*)y.data_ptr(),
                      (const bnbf16*)out.data_ptr(), mean.data_ptr<float>(),
-                     rstd.data_ptr<float>(), s1.data_ptr<float>(),
-                     s2.data_ptr<float>(), C, HW, B, S, (int)act);
+                     rstd.data_ptr<float>(), parts[0].data_ptr<float>(),
+                     parts[1].data_ptr<float>(), C, HW, B, S, (int)act);
+  auto sbuf = parts.sum(1);  // (2, C), contiguous
+  auto s1 = sbuf[0];
+  auto s2 = sbuf[1];
   auto dx = torch::empty_like(y);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(_spatial_chunks(HW), B * C),
                      dim3(256), 0, stream, (const bnbf16*)dy.data_ptr(),

@@ -9,7 +9,8 @@
 //   d qsoft/dx   = sum_l c_l phi_l u'_l - qsoft * sum_l phi_l u'_l,
 //                  u'_l = -2 sigma (x - c_l)
 //   d qsoft/dc_j = phi_j + 2 sigma (x - c_j) phi_j (c_j - qsoft)
-// Gradients for centers are block-reduced then atomically accumulated.
+// Centers gradients: per-wave LDS partials summed in fixed order, plain
+// per-block stores, one ordered host-side sum — bitwise-deterministic.
 //
 // One thread per element; L <= 16 centers live in registers (read through
 // a small constant-ish global array; L=6 in the shipped config).
@@ -102,17 +103,23 @@ __global__ void quantize_bwd_kernel(const float* __restrict__ g,
       gcl[l] += gi * dqc;
     }
   }
-  // block-level reduction of the centers gradient, then one atomic per L
-  __shared__ float red[MAX_L];
-  if (threadIdx.x < MAX_L) red[threadIdx.x] = 0.f;
-  __syncthreads();
+  // DETERMINISTIC centers-grad reduction: per-wave partials in LDS, summed
+  // in fixed wave order by one thread, then a plain per-block store; the
+  // host reduces the (grid, L) partials with one ordered sum. (fp32
+  // atomicAdd — across blocks or across the 4 waves of a block — has a
+  // run-varying order and was one of the two nondeterminism sources.)
+  __shared__ float red[4][MAX_L];
+  const int wv = threadIdx.x >> 6;
 #pragma unroll 8
   for (int l = 0; l < L; ++l) {
     float v = wave_reduce_sum(gcl[l]);
-    if ((threadIdx.x & 63) == 0) atomicAdd(&red[l], v);
+    if ((threadIdx.x & 63) == 0) red[wv][l] = v;
   }
   __syncthreads();
-  if (threadIdx.x < L) atomicAdd(&gc[threadIdx.x], red[threadIdx.x]);
+  if (threadIdx.x < L)
+    gc[(int64_t)blockIdx.x * L + threadIdx.x] =
+        red[0][threadIdx.x] + red[1][threadIdx.x] + red[2][threadIdx.x] +
+        red[3][threadIdx.x];
 }
 
 std::tuple<torch::Tensor, torch::Tensor> quantize_fwd(torch::Tensor x,
@@ -145,16 +152,17 @@ std::tuple<torch::Tensor, torch::Tensor> quantize_bwd(torch::Tensor g,
   CHECK_CUDA_CONTIG(centers);
   int L = centers.numel();
   auto gx = torch::empty_like(x);
-  auto gc = torch::zeros_like(centers);
   int64_t n = x.numel();
   int block = 256;
   int grid = std::min<int64_t>((n + block - 1) / block, 2048);
+  // (grid, L) per-block partials, reduced with one ordered sum
+  auto gcp = torch::empty({grid, L}, centers.options());
   hipLaunchKernelGGL(quantize_bwd_kernel, dim3(grid), dim3(block), 0,
                      at::cuda::getCurrentCUDAStream(),
                      g.data_ptr<float>(), x.data_ptr<float>(),
                      centers.data_ptr<float>(), gx.data_ptr<float>(),
-                     gc.data_ptr<float>(), L, (float)sigma, n);
-  return {gx, gc};
+                     gcp.data_ptr<float>(), L, (float)sigma, n);
+  return {gx, gcp.sum(0)};
 }
 
 }  // namespace dsin

@@ -226,6 +226,42 @@ def test_full_train_step_on_gpu(dev):
         assert torch.isfinite(loss) and torch.isfinite(bpp)
 
 
+def test_step_bitwise_determinism(dev):
+    """Two identical training runs must produce BIT-EQUAL weights: every
+    custom kernel reduces through plain partial stores + ordered sums (no
+    cross-block fp32 atomic accumulation), and the NCC argmax is an
+    order-independent packed-u64 max with index tie-break."""
+    from dsin_amd import config as cm
+    from dsin_amd.models import DSIN
+    from dsin_amd.training import Trainer
+    from dsin_amd.data import SyntheticStereo
+    import os
+    here = os.path.dirname(os.path.abspath(__file__))
+    ae, _ = cm.parse(os.path.join(here, "..", "run_configs", "ae_run_configs"))
+    pc, _ = cm.parse(os.path.join(here, "..", "run_configs", "pc_run_configs"))
+    ae.crop_size = (160, 240)
+
+    def run():
+        torch.manual_seed(33)
+        model = DSIN(ae, pc).to(dev)
+        tr = Trainer(model, ae, pc, num_training_imgs=1576, device=dev,
+                     autocast_bf16=True)
+        gen = SyntheticStereo(160, 240, seed=44, device=str(dev))
+        for _ in range(3):
+            x, y = gen.next_batch()
+            loss, bpp = tr.train_step(x, y)
+        torch.cuda.synchronize()
+        return (float(loss), float(bpp),
+                tr.opt_ae.flat_p.detach().cpu().clone(),
+                tr.opt_pc.flat_p.detach().cpu().clone())
+
+    l1, b1, pa1, pp1 = run()
+    l2, b2, pa2, pp2 = run()
+    assert l1 == l2 and b1 == b2  # bitwise float equality
+    assert torch.equal(pa1, pa2)
+    assert torch.equal(pp1, pp2)
+
+
 # ---------------------------------------------------------------- fused Adam
 
 def test_fused_adam_matches_tf_semantics(dev):
