@@ -25,6 +25,9 @@ def _maybe_int(x: torch.Tensor, cast: bool) -> torch.Tensor:
 
 
 def mae_per_image(x: torch.Tensor, y: torch.Tensor, cast_to_int: bool) -> torch.Tensor:
+    if not cast_to_int and x.is_cuda:
+        from .. import ops
+        return ops.l1_mean_per_image(x, y)  # fused single-pass kernel
     x, y = _maybe_int(x, cast_to_int), _maybe_int(y, cast_to_int)
     return (y - x).abs().float().mean(dim=(1, 2, 3))
 
@@ -97,8 +100,12 @@ def get_loss(config, d_loss_scaled: torch.Tensor, bc: torch.Tensor,
     the reference collects these through tf.losses REGULARIZATION_LOSSES
     (:129-136); we sum them explicitly in the model."""
     assert config.H_target
-    H_real = bc.mean()
-    H_mask = (bc * heatmap).mean() if heatmap is not None else H_real
+    if heatmap is not None:
+        from .. import ops
+        H_real, H_mask = ops.rate_terms(bc, heatmap)  # fused single pass
+    else:
+        H_real = bc.mean()
+        H_mask = H_real
     H_soft = 0.5 * (H_mask + H_real)
     pc_loss = float(config.beta) * torch.clamp(H_soft - float(config.H_target), min=0.0)
     total = d_loss_scaled + pc_loss + reg_loss

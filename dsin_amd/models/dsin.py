@@ -136,7 +136,8 @@ class DSIN(nn.Module):
         bc = self.probclass.bitcost(pc_in.float(), z.symbols, self._pad_value())
         bpp = bitcost_to_bpp(bc, x)
 
-        heatmap = z.heatmap.float() if z.heatmap is not None else None
+        # native dtype: the fused rate_terms kernel reads bf16 heat directly
+        heatmap = z.heatmap if z.heatmap is not None else None
         reg = self.regularization_loss()
         total, H_real, pc_loss = get_loss(
             self.ae_config, (1.0 - self.si_weight) * d.d_loss_scaled, bc, heatmap, reg)
@@ -146,7 +147,8 @@ class DSIN(nn.Module):
         if not self.ae_only:
             assert y is not None and y_dec is not None
             x_with_si = self.side_information(x_dec, y, y_dec)
-            loss_sinet = (x32 - x_with_si.float()).abs().mean()
+            from .. import ops as _ops
+            loss_sinet = _ops.l1_mean_per_image(x32, x_with_si).mean()
 
         loss = total + self.si_weight * loss_sinet
         return {

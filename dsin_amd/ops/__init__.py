@@ -95,15 +95,96 @@ def quantize(x: torch.Tensor, centers: torch.Tensor, sigma: float = 1.0
 
 # ---------------------------------------------------------------------------
 # Heatmap + bottleneck masking (reference src/autoencoder_imgcomp.py:172-201).
-# Small tensors; eager torch math on both devices (uses cuDNN-free pointwise
-# kernels on ROCm — not a custom-kernel hot spot), kept here so models never
-# import `reference` directly.
+# GPU: ONE fused kernel (sigmoid + per-channel clamp ramp + mask multiply)
+# each way instead of the ~5-kernel torch chain per pass.
 # ---------------------------------------------------------------------------
+
+class _HeatmapMaskFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, b: torch.Tensor):
+        z, h3 = _require_ext("heatmap_mask_fwd")(b.contiguous())
+        ctx.save_for_backward(b)
+        ctx.set_materialize_grads(False)
+        return z, h3
+
+    @staticmethod
+    def backward(ctx, gz, gh3):
+        (b,) = ctx.saved_tensors
+        if gz is None:
+            gz = torch.zeros(b.shape[0], b.shape[1] - 1, b.shape[2],
+                             b.shape[3], dtype=b.dtype, device=b.device)
+        gh = gh3.contiguous() if gh3 is not None else None
+        return _require_ext("heatmap_mask_bwd")(b, gz.contiguous(), gh)
+
 
 def heatmap_mask(bottleneck: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     """(N, C+1, H, W) -> (z_masked (N,C,H,W), heatmap3D (N,C,H,W))."""
+    if bottleneck.is_cuda and hip_available():
+        return _HeatmapMaskFn.apply(bottleneck)
     h3 = ref.heatmap3d_ref(bottleneck)
     return h3 * bottleneck[:, 1:], h3
+
+
+# ---------------------------------------------------------------------------
+# Fused loss reductions (SURVEY K10): per-image L1 mean and the rate terms
+# (H_real, H_mask) — single-pass partial sums + ordered host-side reduce,
+# analytic backward kernels. Replaces the sub/abs/mul temporaries and
+# multi-stage torch reduces in the loss assembly.
+# ---------------------------------------------------------------------------
+
+class _L1MeanFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, y: torch.Tensor):
+        x, y = x.contiguous(), y.contiguous()
+        parts = _require_ext("l1_part")(x, y)          # (N, S)
+        chw = x.numel() // x.shape[0]
+        per = parts.sum(dim=1) / float(chw)            # (N,)
+        ctx.save_for_backward(x, y)
+        ctx.chw = chw
+        return per
+
+    @staticmethod
+    def backward(ctx, g):
+        x, y = ctx.saved_tensors
+        gs = (g.float() / float(ctx.chw)).contiguous()
+        gx, gy = _require_ext("l1_bwd")(x, y, gs, ctx.needs_input_grad[0],
+                                        ctx.needs_input_grad[1])
+        return (gx if ctx.needs_input_grad[0] else None,
+                gy if ctx.needs_input_grad[1] else None)
+
+
+def l1_mean_per_image(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    """mean(|y - x|) over CHW per image, shape (N,). GPU fused; CPU eager."""
+    if x.is_cuda and hip_available():
+        return _L1MeanFn.apply(x, y)
+    return (y.float() - x.float()).abs().mean(dim=(1, 2, 3))
+
+
+class _HTermsFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, bc: torch.Tensor, heat: torch.Tensor):
+        bc, heat = bc.contiguous(), heat.contiguous()
+        parts = _require_ext("hterms_part")(bc, heat)  # (S, 2)
+        sums = parts.sum(dim=0) / float(bc.numel())
+        ctx.save_for_backward(bc, heat)
+        return sums[0], sums[1]
+
+    @staticmethod
+    def backward(ctx, g1, g2):
+        bc, heat = ctx.saved_tensors
+        g2v = torch.stack([g1, g2]).float().contiguous()
+        gbc, gheat = _require_ext("hterms_bwd")(bc, heat, g2v,
+                                                ctx.needs_input_grad[1])
+        return gbc, (gheat if ctx.needs_input_grad[1] else None)
+
+
+def rate_terms(bc: torch.Tensor, heat: torch.Tensor):
+    """(mean(bc), mean(bc*heat)) in one pass. GPU fused (bc fp32, heat
+    bf16); eager otherwise."""
+    if (bc.is_cuda and hip_available() and bc.dtype == torch.float32
+            and heat.dtype == torch.bfloat16):
+        return _HTermsFn.apply(bc, heat)
+    return bc.mean(), (bc * heat).mean()
 
 
 # ---------------------------------------------------------------------------

@@ -52,6 +52,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                   torch::Tensor out, torch::Tensor mean,
                                   torch::Tensor rstd, torch::Tensor gamma,
                                   bool training, int64_t act);
+std::vector<torch::Tensor> heatmap_mask_fwd(torch::Tensor b);
+torch::Tensor heatmap_mask_bwd(torch::Tensor b, torch::Tensor gz,
+                               c10::optional<torch::Tensor> gh3);
+torch::Tensor l1_part(torch::Tensor x, torch::Tensor y);
+std::vector<torch::Tensor> l1_bwd(torch::Tensor x, torch::Tensor y,
+                                  torch::Tensor g, bool need_gx,
+                                  bool need_gy);
+torch::Tensor hterms_part(torch::Tensor bc, torch::Tensor heat);
+std::vector<torch::Tensor> hterms_bwd(torch::Tensor bc, torch::Tensor heat,
+                                      torch::Tensor g2, bool need_gheat);
 }  // namespace dsin
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -75,4 +85,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pad_stuff", &dsin::pad_stuff, "fused pad/zero-stuff/cast to bf16");
   m.def("bn_fwd", &dsin::bn_fwd, "fused batch-norm(+act) forward");
   m.def("bn_bwd", &dsin::bn_bwd, "fused batch-norm(+act) backward");
+  m.def("heatmap_mask_fwd", &dsin::heatmap_mask_fwd,
+        "fused heatmap3D + bottleneck mask");
+  m.def("heatmap_mask_bwd", &dsin::heatmap_mask_bwd,
+        "heatmap3D + mask backward");
+  m.def("l1_part", &dsin::l1_part, "per-image |y-x| partial sums");
+  m.def("l1_bwd", &dsin::l1_bwd, "L1-mean backward");
+  m.def("hterms_part", &dsin::hterms_part,
+        "fused (sum bc, sum bc*heatmap) partials");
+  m.def("hterms_bwd", &dsin::hterms_bwd, "rate-term backward");
 }

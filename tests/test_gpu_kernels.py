@@ -226,6 +226,67 @@ def test_full_train_step_on_gpu(dev):
         assert torch.isfinite(loss) and torch.isfinite(bpp)
 
 
+def test_heatmap_mask_matches_eager(dev):
+    from dsin_amd import ops
+    from dsin_amd.ops import reference as ref
+    torch.manual_seed(5)
+    for dtype in (torch.float32, torch.bfloat16):
+        b = torch.randn(2, 33, 10, 12, device=dev, dtype=dtype) * 3
+        b1 = b.clone().requires_grad_(True)
+        z, h3 = ops.heatmap_mask(b1)
+        b2 = b.clone().float().requires_grad_(True)
+        h3r = ref.heatmap3d_ref(b2)
+        zr = h3r * b2[:, 1:]
+        torch.testing.assert_close(z.float(), zr, rtol=2e-2, atol=2e-2)
+        torch.testing.assert_close(h3.float(), h3r, rtol=2e-2, atol=2e-2)
+        gz = torch.randn_like(zr)
+        gh = torch.randn_like(h3r)
+        (z.float() * gz + h3.float() * gh).sum().backward()
+        (zr * gz + h3r * gh).sum().backward()
+        torch.testing.assert_close(b1.grad.float(), b2.grad, rtol=5e-2,
+                                   atol=5e-2)
+
+
+def test_l1_mean_matches_eager(dev):
+    from dsin_amd import ops
+    torch.manual_seed(6)
+    x = torch.randn(3, 4, 33, 47, device=dev) * 50
+    y = (torch.randn(3, 4, 33, 47, device=dev) * 50).to(torch.bfloat16)
+    x1 = x.clone().requires_grad_(True)
+    y1 = y.clone().requires_grad_(True)
+    m = ops.l1_mean_per_image(x1, y1)
+    x2 = x.clone().requires_grad_(True)
+    y2 = y.clone().requires_grad_(True)
+    mr = (y2.float() - x2).abs().mean(dim=(1, 2, 3))
+    torch.testing.assert_close(m, mr, rtol=1e-4, atol=1e-4)
+    g = torch.randn(3, device=dev)
+    (m * g).sum().backward()
+    (mr * g).sum().backward()
+    torch.testing.assert_close(x1.grad, x2.grad, rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(y1.grad.float(), y2.grad.float(), rtol=1e-2,
+                               atol=1e-2)
+
+
+def test_rate_terms_match_eager(dev):
+    from dsin_amd import ops
+    torch.manual_seed(7)
+    bc = (torch.rand(1, 32, 40, 60, device=dev) * 3).requires_grad_(True)
+    heat = torch.rand(1, 32, 40, 60, device=dev).to(torch.bfloat16)
+    heat.requires_grad_(True)
+    hr, hm = ops.rate_terms(bc, heat)
+    bc2 = bc.detach().clone().requires_grad_(True)
+    h2 = heat.detach().clone().requires_grad_(True)
+    hr2 = bc2.mean()
+    hm2 = (bc2 * h2).mean()
+    torch.testing.assert_close(hr, hr2, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(hm, hm2, rtol=1e-3, atol=1e-4)
+    (2.0 * hr + 3.0 * hm).backward()
+    (2.0 * hr2 + 3.0 * hm2).backward()
+    torch.testing.assert_close(bc.grad, bc2.grad, rtol=1e-3, atol=1e-6)
+    torch.testing.assert_close(heat.grad.float(), h2.grad.float(), rtol=1e-2,
+                               atol=1e-5)
+
+
 def test_step_bitwise_determinism(dev):
     """Two identical training runs must produce BIT-EQUAL weights: every
     custom kernel reduces through plain partial stores + ordered sums (no
