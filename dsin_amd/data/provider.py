@@ -171,10 +171,19 @@ class SyntheticStereo:
         img = torch.nn.functional.interpolate(
             base, size=(h, w + 64), mode="bilinear", align_corners=False)
         img = img * 220.0 + torch.rand(n, 3, h, w + 64, generator=g) * 35.0
+        # standardize each image to fixed KITTI-like global statistics:
+        # real KITTI has stable per-image stats (that is what makes the
+        # reference's FIXED normalization work, src/AE.py:222-248); without
+        # this the raw field's per-image mean/contrast swing makes batch-1
+        # BN behave like instance norm at train but mismatch its running
+        # stats at eval, inflating eval-mode bpp on synthetic data
+        mu = img.mean(dim=(2, 3), keepdim=True)
+        sd = img.std(dim=(2, 3), keepdim=True).clamp_min(1e-3)
+        img = (img - mu) / sd * 70.0 + 96.0
         shift = int(torch.randint(4, 48, (1,), generator=g))
         x = img[..., 64:64 + w]
         y = img[..., 64 - shift:64 - shift + w].clone()
-        y += torch.randn(n, 3, h, w, generator=g) * 2.0
+        y = y + torch.randn(n, 3, h, w, generator=g) * 2.0
         x = x.clamp(0, 255).to(self.device)
         y = y.clamp(0, 255).to(self.device)
         return x, y

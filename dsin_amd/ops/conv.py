@@ -54,22 +54,36 @@ _WMAT_EPOCH = 0
 
 
 def begin_step() -> None:
-    """Invalidate the per-step W-panel cache (call when weights change)."""
+    """Invalidate the per-step W-panel cache (call when weights change).
+    Entries are only ever valid within their own epoch, so clearing is
+    free — and it releases the pinned source tensors (see _wmat_cache)."""
     global _WMAT_EPOCH
     _WMAT_EPOCH += 1
-    if len(_WMAT_CACHE) > 4096:  # unbounded-growth guard (tests, many shapes)
-        _WMAT_CACHE.clear()
+    _WMAT_CACHE.clear()
 
 
 def _wmat_cache(w1: torch.Tensor, tag, builder):
+    """Cache a built panel for w1 under (data_ptr, shape, tag) for the
+    current epoch. The entry PINS w1 itself: several call sites pass
+    TEMPORARIES (probclass weight*mask products, conv-transpose
+    flip/permute chains) whose storage would otherwise be freed and
+    recycled by the caching allocator — a later conv's lookup could then
+    hit this entry's key with a DIFFERENT weight at the same address and
+    silently run with another layer's panel. (Exactly that happened under
+    no_grad, where autograd keeps no references: every eval-mode forward
+    ran the probclass with shuffled panels while training looked fine.)
+    Holding the reference makes live keys unique by construction."""
     if not w1.is_cuda:
         return builder()
     key = (w1.data_ptr(), tuple(w1.shape), tag)
     ent = _WMAT_CACHE.get(key)
     if ent is not None and ent[0] == _WMAT_EPOCH:
+        # safe: ent pins its source tensor, so a live entry's address
+        # cannot have been recycled — a key match means w1 views the very
+        # same (unmodified-this-epoch) storage
         return ent[1]
     panel = builder()
-    _WMAT_CACHE[key] = (_WMAT_EPOCH, panel)
+    _WMAT_CACHE[key] = (_WMAT_EPOCH, panel, w1)
     return panel
 
 

@@ -287,6 +287,40 @@ def test_rate_terms_match_eager(dev):
                                atol=1e-5)
 
 
+def test_forward_grad_nograd_agree(dev):
+    """A no_grad forward must equal the grad-mode forward. Regression: the
+    W-panel cache keyed on data_ptr; under no_grad the probclass
+    weight*mask and conv-transpose flip temporaries were freed instantly,
+    the allocator recycled their addresses, and a later conv silently ran
+    with ANOTHER layer's cached panel (eval-mode bpp ~3x off while
+    training looked perfect)."""
+    import os
+    from dsin_amd import config as cm
+    from dsin_amd.models import DSIN
+    from dsin_amd.training import Trainer
+    from dsin_amd.data import SyntheticStereo
+    here = os.path.dirname(os.path.abspath(__file__))
+    ae, _ = cm.parse(os.path.join(here, "..", "run_configs", "ae_run_configs"))
+    pc, _ = cm.parse(os.path.join(here, "..", "run_configs", "pc_run_configs"))
+    ae.crop_size = (160, 240)
+    torch.manual_seed(3)
+    model = DSIN(ae, pc).to(dev)
+    tr = Trainer(model, ae, pc, 1576, device=dev, autocast_bf16=True)
+    gen = SyntheticStereo(160, 240, seed=8, device=str(dev))
+    for _ in range(3):
+        x, y = gen.next_batch()
+        tr.train_step(x, y)
+    outs = []
+    for grad in (True, False, True):
+        ctx = (torch.enable_grad() if grad else torch.no_grad())
+        with ctx, tr._autocast():
+            o = model.train_losses(x, y)
+        outs.append((float(o["loss"]), float(o["bpp"])))
+    (l0, b0), (l1, b1), (l2, b2) = outs
+    assert abs(b1 - b0) < 1e-3 and abs(b2 - b0) < 1e-3, outs
+    assert abs(l1 - l0) / max(abs(l0), 1.0) < 1e-2, outs
+
+
 def test_step_bitwise_determinism(dev):
     """Two identical training runs must produce BIT-EQUAL weights: every
     custom kernel reduces through plain partial stores + ordered sums (no

@@ -61,16 +61,13 @@ def main():
         t_mi_f = timeit(lambda: F.conv2d(xb, wb, None, stride=st, padding=pad,
                                          dilation=d), args.iters)
 
-        # wrw alone
-        xr = x.requires_grad_(False)
+        # wrw via autograd (same route the training step takes)
         dy = torch.randn(1, Co, HO, WO, device=dev).to(torch.bfloat16)
-        from dsin_amd.ops.conv import _plan
-        from dsin_amd.ops import _require_ext
-        ext_wrw = _require_ext("conv_wrw")
-        xbuf = F.pad(x.to(torch.bfloat16), (pad,) * 4).contiguous()
-        mb, ko = _plan(dev, Ci, H + 2 * pad, W + 2 * pad, k, k, st, d, HO, WO)
-        t_ours_w = timeit(lambda: ext_wrw(xbuf, dy, mb, ko, Co, Ci * k * k,
-                                          WO, st == 1), args.iters)
+        w2 = w.clone().requires_grad_(True)
+        def ours_w():
+            yv = dconv.conv2d(x.detach(), w2, None, st, pad, d)
+            torch.autograd.grad(yv, w2, dy.float())
+        t_ours_w = timeit(ours_w, args.iters)
         wg = wb.requires_grad_(True)
         def mi_w():
             y = F.conv2d(xb, wg, None, stride=st, padding=pad, dilation=d)
