@@ -52,6 +52,9 @@ _COMPUTE = "bf16"
 _WMAT_CACHE = {}
 _WMAT_EPOCH = 0
 
+# gate for the direct 5x5/stride-2 LDS-halo forward kernel
+_DIRECT5 = True
+
 
 def begin_step() -> None:
     """Invalidate the per-step W-panel cache (call when weights change).
@@ -179,10 +182,12 @@ def _dummy_tables(device):
     return p
 
 
-def _wmat(w1: torch.Tensor, fp8: bool = False,
-          direct: bool = False) -> torch.Tensor:
+def _wmat(w1: torch.Tensor, fp8: bool = False, direct: bool = False,
+          khw: int = 9) -> torch.Tensor:
     """(Co, K) any-dtype -> bf16 (or e4m3-as-uint8) zero-padded to
-    (Co, KP64+8); the zeros cancel the clamped out-of-range A gathers."""
+    (Co, KP64+8); the zeros cancel the clamped out-of-range A gathers.
+    direct: (chunk, tap, ci) mode-2 layout for the direct LDS-halo kernels
+    (khw = 9 for 3x3, 25 for 5x5/s2)."""
     K = w1.shape[1]
     if fp8:
         def build8():
@@ -195,9 +200,9 @@ def _wmat(w1: torch.Tensor, fp8: bool = False,
         return _wmat_cache(w1, "fp8", build8)
     if w1.is_cuda and hip_available():
         return _wmat_cache(
-            w1, "d" if direct else "p",
+            w1, ("d", khw) if direct else "p",
             lambda: _require_ext("wmat_make")(w1.contiguous(),
-                                              9 if direct else 1,
+                                              khw if direct else 1,
                                               2 if direct else 0))
     KP = (K + 63) & ~63
     return F.pad(w1.to(torch.bfloat16), (0, KP + 8 - K)).contiguous()
@@ -243,7 +248,12 @@ class _GatherConvFn(torch.autograd.Function):
         B, Ci, H, W = x.shape
         Co, K = w1.shape
         bias32 = bias.float().contiguous() if bias is not None else None
-        if direct:
+        if direct == 2:
+            dt = _dummy_tables(x.device)
+            y = ext_fwd(x, _wmat(w1, direct=True, khw=25), bias32, dt[0],
+                        dt[1], Co, K, HO, WO, act, 2, 2, pt, 0, 0, 0, 1,
+                        None, 0, 0, 1, WO)
+        elif direct:
             dt = _dummy_tables(x.device)
             y = ext_fwd(x, _wmat(w1, direct=True), bias32, dt[0], dt[1], Co,
                         K, HO, WO, act, 1, 1, pt, 0, 0, 0, 1,
@@ -435,8 +445,13 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
                                       stride, dilation, kh, kw, HO, WO, act,
                                       (padding, padding, padding, padding), 1)
     x = x.to(torch.bfloat16).contiguous()
-    direct = (stride == 1 and dilation == 1 and kh == 3 and kw == 3
-              and Ci % 64 == 0 and padding >= 1)
+    direct = 0
+    if stride == 1 and dilation == 1 and kh == 3 and kw == 3 \
+            and Ci % 64 == 0 and padding >= 1:
+        direct = 1
+    elif stride == 2 and dilation == 1 and kh == 5 and kw == 5 \
+            and Ci % 64 == 0 and _DIRECT5:
+        direct = 2
     return _GatherConvFn.apply(x, w.reshape(Co, Ci * kh * kw), bias,
                                stride, 1, dilation, kh, kw, padding, padding,
                                HO, WO, act, direct)

@@ -54,6 +54,35 @@ torch::Tensor conv_fwd(torch::Tensor xbuf, torch::Tensor wmat,
   TORCH_CHECK(wmat.scalar_type() == xbuf.scalar_type(), "x/w dtype mismatch");
   const int64_t B = xbuf.size(0);
   const int64_t M = HO * WO;
+  if (direct == 2 && !fp8) {
+    // 5x5 stride-2 direct LDS-halo kernel (virtual pad)
+    const int64_t Ci = xbuf.size(1);
+    const int64_t KPd = (K + 63) & ~63;
+    TORCH_CHECK(K == Ci * 25 && Ci % 64 == 0 && stride == 2,
+                "direct5 conv gate mismatch");
+    TORCH_CHECK(wmat.size(1) == KPd + CONV_AP, "direct5 wmat stride mismatch");
+    const int Hp = (int)xbuf.size(2), Wp = (int)xbuf.size(3);
+    TORCH_CHECK(HO == (Hp + 2 * vpad - 5) / 2 + 1 &&
+                    WO == (Wp + 2 * vpad - 5) / 2 + 1,
+                "direct5 conv size mismatch");
+    auto out = torch::empty({B, N, HO, WO},
+                            xbuf.options().dtype(torch::kBFloat16));
+    const float* bp = nullptr;
+    if (bias.has_value()) {
+      CHECK_CUDA_CONTIG(bias.value());
+      bp = bias->data_ptr<float>();
+    }
+    dim3 grid(((HO + 7) / 8) * ((WO + 7) / 8), (N + 63) / 64, B);
+    size_t lds = (size_t)19 * 19 * 64 * 2;
+    hipLaunchKernelGGL(conv5x5s2_direct_kernel, grid, dim3(256), lds,
+                       at::cuda::getCurrentCUDAStream(),
+                       (const cvbf16*)xbuf.data_ptr(),
+                       (const cvbf16*)wmat.data_ptr(), bp,
+                       (cvbf16*)out.data_ptr(), (int)Ci, Hp, Wp, (int)N,
+                       (int)HO, (int)WO, (int)KPd, xbuf.stride(0),
+                       (long long)N * M, (int)act, (int)vpad);
+    return out;
+  }
   if (direct && !fp8) {
     // stride-1 3x3, Ci%64==0: direct LDS-halo kernel (9x less gather
     // traffic than the im2col path); caller built wmat in mode-2/3 layout

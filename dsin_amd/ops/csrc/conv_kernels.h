@@ -687,6 +687,164 @@ void conv3x3_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
   }
 }
 
+// Direct 5x5 stride-2 convolution (encoder h2 / to_bn shapes, Ci % 64 ==
+// 0). The gather path re-reads each input element 25x through L2 and
+// discards half of every stride-2 staging vector; here each 8x8-output
+// tile's 19x19 input halo is staged ONCE per 64-channel chunk (single
+// 46 KB LDS buffer -> 3 workgroups/CU) and all 25 taps run out of LDS:
+// 200 MFMAs between barriers per wave. W panel layout (chunk, tap, ci) =
+// wmat_make mode 2 with khw = 25. Always VIRTUAL pad (vp = padding).
+__global__ __launch_bounds__(256)
+void conv5x5s2_direct_kernel(const cvbf16* __restrict__ xpad, // (Ci, Hp, Wp)
+                             const cvbf16* __restrict__ wmat, // (N, KP+AP)
+                             const float* __restrict__ bias,
+                             cvbf16* __restrict__ out,        // (N, HO, WO)
+                             int Ci, int Hp, int Wp, int N, int HO, int WO,
+                             int KP, long long x_img_stride,
+                             long long o_img_stride, int act, int vp) {
+  constexpr int CIC = 64;
+  constexpr int XT = 19;             // odd row stride breaks 2-row aliasing
+  constexpr int TB = 19 * XT * CIC;  // bf16 elems, single buffer
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* As8 = smem;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int tx = (WO + 7) >> 3;
+  const int oy0 = (blockIdx.x / tx) * 8;
+  const int ox0 = (blockIdx.x % tx) * 8;
+  const long long img = blockIdx.z;
+  const cvbf16* x = xpad + img * x_img_stride;
+
+  const int colL = lane & 15;
+  const int kgrp = lane >> 4;
+  const int n0 = blockIdx.y * CONV_TN + wid * 16;
+  const int ncol = n0 + colL;
+  const int WSTRIDE = KP + CONV_AP;
+  const cvbf16* wrow = wmat + (long long)(ncol < N ? ncol : 0) * WSTRIDE;
+
+  cv_f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                     {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+
+  // 16B-granule XOR swizzle. The tap loop reads pixels at STRIDE 2 (256 B
+  // = a full bank row), so the plain (p & 7) key of the 3x3 kernel only
+  // takes 4 values across the 8 px lanes; folding p's low bit into bit 2
+  // of the key gives 8 distinct granules for both parities.
+  auto aoff = [&](int p, int ci) -> int {
+    const int key = ((p >> 1) & 7) ^ ((p & 1) << 2);
+    return p * (CIC * 2) + ((((ci >> 3) ^ key) << 4) | ((ci & 7) << 1));
+  };
+
+  typedef __attribute__((ext_vector_type(8))) unsigned short u16x8;
+  const int gy0 = 2 * oy0 - vp;
+  const int cbase = 2 * ox0 - vp;
+  const bool tile_int = gy0 >= 0 && gy0 + 19 <= Hp && cbase >= 0 &&
+                        cbase + 24 <= Wp;
+  auto stage = [&](int c) {
+    for (int ridx = tid; ridx < CIC * 19; ridx += 256) {
+      const int ci = ridx / 19, y = ridx % 19;
+      if (tile_int) {
+        const cvbf16* g =
+            x + ((long long)(c * CIC + ci) * Hp + (gy0 + y)) * Wp + cbase;
+        const u16x8 a = *reinterpret_cast<const u16x8*>(g);
+        const u16x8 b = *reinterpret_cast<const u16x8*>(g + 8);
+        const u16x8 e = *reinterpret_cast<const u16x8*>(g + 16);
+#pragma unroll
+        for (int xi = 0; xi < 19; ++xi) {
+          unsigned short v =
+              xi < 8 ? a[xi] : (xi < 16 ? b[xi - 8] : e[xi - 16]);
+          *reinterpret_cast<cvbf16*>(&As8[aoff(y * XT + xi, ci)]) =
+              *reinterpret_cast<cvbf16*>(&v);
+        }
+      } else {  // border: clamped unconditional loads + select
+        const int gy = gy0 + y;
+        const bool rowin = gy >= 0 && gy < Hp;
+        const long long rowb =
+            ((long long)(c * CIC + ci) * Hp + (rowin ? gy : 0)) * Wp;
+#pragma unroll
+        for (int xi = 0; xi < 19; ++xi) {
+          const int col = cbase + xi;
+          const cvbf16 lv = x[rowb + min(max(col, 0), Wp - 1)];
+          const cvbf16 v =
+              (rowin && (unsigned)col < (unsigned)Wp) ? lv : cvf2b(0.f);
+          *reinterpret_cast<cvbf16*>(&As8[aoff(y * XT + xi, ci)]) = v;
+        }
+      }
+    }
+  };
+
+  const int nchunks = Ci / CIC;
+  cv_bf16x8 wA[2], wB[2];
+  auto load_w = [&](int kbase, cv_bf16x8* wset) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+      wset[j] = *reinterpret_cast<const cv_bf16x8*>(
+          &wrow[kbase + j * 32 + kgrp * 8]);
+  };
+
+  stage(0);
+  load_w(0, wA);
+  __syncthreads();
+
+  auto run_chunk = [&](auto codd, int c) {
+    constexpr int CO = decltype(codd)::value;  // (c*25)&1 == c&1 (25 odd)
+    const int kb = c * 25 * CIC;
+#pragma unroll
+    for (int tap = 0; tap < 25; ++tap) {
+      cv_bf16x8* wcur = ((CO + tap) & 1) ? wB : wA;
+      cv_bf16x8* wnxt = ((CO + tap) & 1) ? wA : wB;
+      if (tap < 24) load_w(kb + (tap + 1) * CIC, wnxt);
+      else if (c + 1 < nchunks) load_w(kb + 25 * CIC, wnxt);
+      const int r = tap / 5, sx = tap % 5;
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int cio = (kk << 5) + kgrp * 8;
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          const int m = mi * 16 + colL;
+          const int p = ((m >> 3) * 2 + r) * XT + (m & 7) * 2 + sx;
+          const cv_bf16x8 afrag = *reinterpret_cast<const cv_bf16x8*>(
+              &As8[aoff(p, cio)]);
+          acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, wcur[kk], acc[mi], 0, 0, 0);
+        }
+      }
+    }
+  };
+  constexpr std::integral_constant<int, 0> D0{};
+  constexpr std::integral_constant<int, 1> D1{};
+  for (int c = 0; c < nchunks; ++c) {
+    if (c & 1) run_chunk(D1, c);
+    else run_chunk(D0, c);
+    if (c + 1 < nchunks) {
+      __syncthreads();   // all waves done reading the single buffer
+      stage(c + 1);
+      __syncthreads();
+    }
+  }
+
+  const float bv = (bias != nullptr && ncol < N) ? bias[ncol] : 0.f;
+  cvbf16* o = out + img * o_img_stride +
+              (long long)(ncol < N ? ncol : 0) * HO * WO;
+  if (ncol < N) {
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int m = mi * 16 + kgrp * 4 + reg;
+        const int oy = oy0 + (m >> 3), ox = ox0 + (m & 7);
+        if (oy < HO && ox < WO) {
+          float v = acc[mi][reg] + bv;
+          if (act == 1) v = fmaxf(v, 0.f);
+          else if (act == 2) v = fmaxf(v, 0.2f * v);
+          o[(long long)oy * WO + ox] = cvf2b(v);
+        }
+      }
+    }
+  }
+}
+
 // dW[co][k] += sum over the workgroup's pixel chunk of dy[co][m]*A[m][k].
 // GEMM roles: A' (M'=filter taps K) gathered rows, B' = dy columns.
 // Tile: M'64 (taps) x N'64 (couts), K' = pixels chunked by 32.
