@@ -183,6 +183,29 @@ def _run_trainstep_world(world, fused):
         assert abs(results[0][1] - results[r][1]).max() < 1e-6
 
 
+def test_flat_grad_reducer_eight_ranks():
+    """8-rank flat-reducer equivalence (the driver's max GPU count)."""
+    import socket
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        port = _s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    world = 8
+    procs = [ctx.Process(target=_worker_flat_reducer, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, flat_g, x, slices = q.get(timeout=300)
+        results[rank] = flat_g
+    for p in procs:
+        p.join(timeout=60)
+    for r in range(1, world):
+        assert abs(results[0] - results[r]).max() < 1e-6
+
+
 def test_full_train_step_replicas_stay_synced():
     _run_trainstep_world(2, fused=None)
 

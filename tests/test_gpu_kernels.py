@@ -287,6 +287,26 @@ def test_rate_terms_match_eager(dev):
                                atol=1e-5)
 
 
+def test_codec_roundtrip_on_gpu(dev):
+    """Wavefront entropy codec end-to-end on the GPU kernels: encoder and
+    decoder run identical per-wave batched network calls, so the roundtrip
+    must be bit-exact (guaranteed by the now-bitwise-deterministic conv3d
+    path)."""
+    from dsin_amd import config as cm
+    from dsin_amd.coding import decode_symbols, encode_symbols
+    from dsin_amd.models.probclass import ProbClass
+    import os
+    here = os.path.dirname(os.path.abspath(__file__))
+    pcc, _ = cm.parse(os.path.join(here, "..", "run_configs", "pc_run_configs"))
+    torch.manual_seed(0)
+    pc = ProbClass(pcc, num_centers=6).to(dev)
+    centers = torch.linspace(-2, 2, 6, device=dev)
+    symbols = torch.randint(0, 6, (4, 8, 10), device=dev)
+    data = encode_symbols(pc, centers, symbols)
+    out = decode_symbols(pc, centers, data, (4, 8, 10))
+    assert torch.equal(out.cpu(), symbols.cpu())
+
+
 def test_forward_grad_nograd_agree(dev):
     """A no_grad forward must equal the grad-mode forward. Regression: the
     W-panel cache keyed on data_ptr; under no_grad the probclass

@@ -32,7 +32,7 @@ def test_main_cli_end_to_end(tmp_path):
         [sys.executable, os.path.join(REPO, "main.py"),
          "-ae_config", os.path.join(d, "run_configs", "ae_run_configs"),
          "-pc_config", os.path.join(d, "run_configs", "pc_run_configs"),
-         "--metrics", os.path.join(d, "m.jsonl")],
+         "--metrics", os.path.join(d, "m.jsonl"), "--plots"],
         cwd=d, env=env, capture_output=True, text=True, timeout=900)
     assert r.returncode == 0, r.stderr[-2000:]
     weights = os.listdir(os.path.join(d, "weights"))
@@ -48,3 +48,8 @@ def test_main_cli_end_to_end(tmp_path):
     assert {n.split("_")[0] for n in lists} >= {"bpp", "l1", "psnr", "mse",
                                                "msssim", "pearson"}
     assert os.path.exists(os.path.join(d, "m.jsonl"))
+    # --plots wrote the loss figure and the 5-panel inference figure
+    figs = [f for f in os.listdir(os.path.join(d, "images"))
+            if f.endswith(".png")]
+    assert any(f.startswith("loss_") for f in figs), figs
+    assert any(f.startswith("inference_") for f in figs), figs
