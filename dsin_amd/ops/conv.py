@@ -260,11 +260,14 @@ class _GatherConvFn(torch.autograd.Function):
                         None, 0, 0, 1, WO)
         elif sv == 2 and dil == 1:
             # conv-transpose: 4 dense stride-1 phase sub-convs (see
-            # _phase_plans) writing strided into one full-grid output
+            # _phase_plans) writing strided into one full-grid output.
+            # A phase with no matching taps (k=1 dims) is mathematically
+            # zero but never written -> zero-init unless all 4 phases run.
             pg = _require_ext("panel_gather")
-            y = torch.empty(B, Co, HO, WO, dtype=torch.bfloat16,
-                            device=x.device)
-            for p_ in _phase_plans(x.device, HO, WO, kh, kw, Ci, pt, pl):
+            plans = _phase_plans(x.device, HO, WO, kh, kw, Ci, pt, pl)
+            alloc = torch.empty if len(plans) == 4 else torch.zeros
+            y = alloc(B, Co, HO, WO, dtype=torch.bfloat16, device=x.device)
+            for p_ in plans:
                 wp = _wmat_cache(
                     w1, ("ph", p_["a"], p_["b"], kh, kw),
                     lambda p_=p_: pg(w1.contiguous(), p_["ktab"]))
@@ -307,10 +310,12 @@ class _GatherConvFn(torch.autograd.Function):
                     # with per-phase column-gathers of the rotated panel
                     pg = _require_ext("panel_gather")
                     rot = _wmat_rot(w1, kh * kw)
-                    dx = torch.empty(B, Ci, H, W, dtype=torch.bfloat16,
-                                     device=x.device)
-                    for p_ in _phase_plans(dy.device, H, W, kh, kw, Co,
-                                           ptb, plb):
+                    plans_b = _phase_plans(dy.device, H, W, kh, kw, Co,
+                                           ptb, plb)
+                    alloc = torch.empty if len(plans_b) == 4 else torch.zeros
+                    dx = alloc(B, Ci, H, W, dtype=torch.bfloat16,
+                               device=x.device)
+                    for p_ in plans_b:
                         wp = _wmat_cache(
                             w1, ("phr", p_["a"], p_["b"], kh, kw),
                             lambda p_=p_: pg(rot, p_["ktab"]))

@@ -10,7 +10,6 @@ from __future__ import annotations
 import argparse
 import datetime
 import os
-import sys
 
 import numpy as np
 import torch
