@@ -68,3 +68,52 @@ def test_load_for_si_inference(small_ae_config, pc_config, tmp_path):
     step = checkpoint.load(m2, tr2.optimizers, os.path.join(root, "m"), cfg)
     assert step == 0
     assert torch.equal(m2.sinet.last.weight, m1.sinet.last.weight)
+
+
+def test_resume_equivalence(small_ae_config, pc_config, tmp_path):
+    """Crash-recovery semantics: train 2 steps, checkpoint with training
+    state, resume in a fresh process-equivalent Trainer, train 2 more —
+    must equal 4 uninterrupted steps (weights AND optimizer state). Uses
+    the fused flat-buffer Adam (the GPU path) via its CPU fallback."""
+    from dsin_amd.data import SyntheticStereo
+    root = str(tmp_path) + os.sep
+    cfg = small_ae_config
+    gen = lambda: SyntheticStereo(cfg.crop_size[0], cfg.crop_size[1], seed=11)
+
+    def steps(tr, g, n):
+        out = None
+        for _ in range(n):
+            x, y = g.next_batch()
+            out = tr.train_step(x, y)
+        return out
+
+    # uninterrupted run
+    m1 = make_model(cfg, pc_config, 7)
+    tr1 = Trainer(m1, cfg, pc_config, num_training_imgs=10, fused_adam=True)
+    g1 = gen()
+    steps(tr1, g1, 4)
+
+    # interrupted + resumed run
+    m2 = make_model(cfg, pc_config, 7)
+    tr2 = Trainer(m2, cfg, pc_config, num_training_imgs=10, fused_adam=True)
+    g2 = gen()
+    steps(tr2, g2, 2)
+    checkpoint.save(m2, tr2.optimizers, tr2.global_step, root, "resume",
+                    2, 4, 9.9, cfg, pc_config)
+    lcfg = cfg.clone(load_train_step=True)
+    m4 = make_model(cfg, pc_config, 8)   # different init; load must fix
+    tr4 = Trainer(m4, cfg, pc_config, num_training_imgs=10, fused_adam=True)
+    step = checkpoint.load(m4, tr4.optimizers, os.path.join(root, "resume"),
+                           lcfg)
+    tr4.global_step = step
+    g4 = gen()
+    for _ in range(2):      # skip the two already-trained batches
+        g4.next_batch()
+    steps(tr4, g4, 2)
+
+    torch.testing.assert_close(tr4.opt_ae.flat_p, tr1.opt_ae.flat_p,
+                               rtol=0, atol=0)
+    torch.testing.assert_close(tr4.opt_pc.flat_p, tr1.opt_pc.flat_p,
+                               rtol=0, atol=0)
+    torch.testing.assert_close(tr4.opt_ae.exp_avg, tr1.opt_ae.exp_avg,
+                               rtol=0, atol=0)
