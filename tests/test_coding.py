@@ -47,7 +47,7 @@ def test_probclass_testing_matches_bitcost(pc_config):
     total = pt.total_bit_cost(symbols)
     q = centers[symbols.unsqueeze(0)]
     bc = pc.bitcost(q, symbols.unsqueeze(0), centers[0])
-    assert abs(total - float(bc.sum())) < 1e-3
+    assert abs(total - float(bc.detach().sum())) < 1e-3
 
 
 def test_prediction_network_shapes(pc_config):
