@@ -18,6 +18,7 @@ replays. Capture failure falls back to eager with a warning.
 from __future__ import annotations
 
 import contextlib
+import os
 import warnings
 from typing import Optional
 
@@ -78,7 +79,9 @@ class Trainer:
                  graph_warmup: int = 3,
                  ddp_bucket_bytes: int = 8 * 1024 * 1024,
                  ddp_comm_dtype: Optional[torch.dtype] = None,
-                 fused_adam: Optional[bool] = None):
+                 fused_adam: Optional[bool] = None,
+                 nan_guard: Optional[bool] = None,
+                 blackbox_dir: str = "nan_blackbox"):
         self.model = model
         self.ae_config = ae_config
         self.device = device or next(model.parameters()).device
@@ -128,6 +131,17 @@ class Trainer:
         self._static_x = None
         self._static_y = None
         self._static_out = None
+
+        # failure detection: opt-in per-step
+        # non-finite guard that dumps a "black box" (failing batch, model +
+        # optimizer state, per-component losses, per-buffer corruption map)
+        # and raises — the diagnostic for the open RD-run stability item
+        # (profiles/r02_rd_curve.md). Costs one host sync per step, so it is
+        # off by default; enable with nan_guard=True or DSIN_NANCHECK=1.
+        if nan_guard is None:
+            nan_guard = os.environ.get("DSIN_NANCHECK", "0") == "1"
+        self.nan_guard = bool(nan_guard)
+        self.blackbox_dir = blackbox_dir
 
     @property
     def optimizers(self):
@@ -230,7 +244,8 @@ class Trainer:
             self._graph = None
 
     def train_step(self, x: torch.Tensor, y: Optional[torch.Tensor]):
-        """Returns (loss, bpp) tensors on device (no host sync)."""
+        """Returns (loss, bpp) tensors on device (no host sync unless
+        nan_guard is enabled)."""
         self.sched_ae.set_step(self.global_step)
         self.sched_pc.set_step(self.global_step)
 
@@ -240,6 +255,8 @@ class Trainer:
                 self._static_y.copy_(y, non_blocking=True)
             self._graph.replay()
             self.global_step += 1
+            if self.nan_guard:
+                self._check_finite(self._static_out[0], x, y)
             return self._static_out
 
         if (self.use_cuda_graph and not self._graph_failed
@@ -253,11 +270,78 @@ class Trainer:
                     self._static_y.copy_(y, non_blocking=True)
                 self._graph.replay()
                 self.global_step += 1
+                if self.nan_guard:
+                    self._check_finite(self._static_out[0], x, y)
                 return self._static_out
 
         out = self._step_inner(x, y)
         self.global_step += 1
+        if self.nan_guard:
+            self._check_finite(out[0], x, y)
         return out
+
+    # -- failure detection --------------------------------------------------
+
+    def _check_finite(self, loss: torch.Tensor, x, y) -> None:
+        if bool(torch.isfinite(loss)):
+            return
+        path = self._dump_blackbox(x, y)
+        raise RuntimeError(
+            f"non-finite training loss at global step {self.global_step}; "
+            f"black box saved to {path}")
+
+    @staticmethod
+    def _buf_stats(t: Optional[torch.Tensor]) -> dict:
+        if t is None or not isinstance(t, torch.Tensor) or t.numel() == 0:
+            return {}
+        tf = t.detach().float()
+        bad = (~torch.isfinite(tf)).sum()
+        return {"numel": t.numel(), "nonfinite": int(bad),
+                "absmax": float(tf.nan_to_num_(posinf=0, neginf=0).abs().max())}
+
+    @torch.no_grad()
+    def _dump_blackbox(self, x, y) -> str:
+        """Everything needed to replay the failing step offline: the batch,
+        full model + optimizer state (post-step, i.e. possibly already
+        corrupt), per-component loss values from an eager re-run, and a
+        per-buffer corruption map locating the first non-finite storage."""
+        os.makedirs(self.blackbox_dir, exist_ok=True)
+        step = self.global_step
+        components = {}
+        try:
+            with self._autocast():
+                out = self.model.train_losses(x, y)
+            for k in ("loss", "bpp", "H_real", "pc_loss", "d_loss", "reg",
+                      "loss_sinet"):
+                v = out.get(k)
+                if isinstance(v, torch.Tensor) and v.numel() == 1:
+                    components[k] = float(v.detach())
+        except Exception as e:  # the re-run itself may blow up — still dump
+            components["rerun_error"] = repr(e)
+        param_map = {n: self._buf_stats(p)
+                     for n, p in self.model.named_parameters()}
+        opt_map = {}
+        for tag, opt in (("ae", self.opt_ae), ("pc", self.opt_pc)):
+            for buf in ("flat_p", "flat_g", "exp_avg", "exp_avg_sq"):
+                opt_map[f"{tag}.{buf}"] = self._buf_stats(
+                    getattr(opt, buf, None))
+        path = os.path.join(self.blackbox_dir, f"step_{step}.pt")
+        torch.save({
+            "global_step": step,
+            "x": x.detach().cpu(), "y": None if y is None else y.detach().cpu(),
+            "components": components,
+            "param_stats": param_map,
+            "optimizer_stats": opt_map,
+            "model_state": {k: v.cpu() for k, v in
+                            self.model.state_dict().items()},
+            "optim_state": [opt.state_dict() for opt in self.optimizers],
+        }, path)
+        corrupt = {k: v for k, v in {**param_map, **opt_map}.items()
+                   if v.get("nonfinite")}
+        warnings.warn(f"nan_guard tripped at step {step}: components="
+                      f"{components}; corrupt buffers: "
+                      f"{list(corrupt)[:8] or 'none (transient activations)'}")
+        return path
 
     def validate(self, x: torch.Tensor, y: Optional[torch.Tensor]) -> torch.Tensor:
         from ..ops import conv as _conv

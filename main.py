@@ -52,6 +52,10 @@ def main(argv=None):
                     help="periodic crash-recovery checkpoint interval "
                          "(iterations; 0 = best-val only, the reference's "
                          "behavior)")
+    ap.add_argument("--nan-guard", action="store_true",
+                    help="per-step non-finite loss check; on failure dumps "
+                         "batch + model/optimizer state + corruption map to "
+                         "nan_blackbox/ and aborts (also: DSIN_NANCHECK=1)")
     ap.add_argument("--plots", action="store_true",
                     help="save loss / inference figures (headless "
                          "equivalents of the reference's plt.show windows, "
@@ -74,7 +78,8 @@ def main(argv=None):
     data = make_dataset(ae_config, data_dir, seed=rank())
     num_train = len(data.train_pairs)
     trainer = Trainer(model, ae_config, pc_config, num_train, device=device,
-                     autocast_bf16=args.bf16, use_cuda_graph=args.graphs)
+                     autocast_bf16=args.bf16, use_cuda_graph=args.graphs,
+                     nan_guard=args.nan_guard or None)
     metrics = MetricsLogger(args.metrics if rank() == 0 else None)
 
     model_name = "NA"
