@@ -121,3 +121,43 @@ def test_codec_bits_close_to_bitcost(pc_config):
     actual_bits = len(data) * 8
     # coder overhead: freq quantization + 4 flush bytes
     assert actual_bits < est_bits * 1.05 + 64, (actual_bits, est_bits)
+
+
+try:
+    from hypothesis import given, settings, strategies as st
+    HAVE_HYP = True
+except ImportError:  # pragma: no cover
+    HAVE_HYP = False
+
+
+if HAVE_HYP:
+    @settings(max_examples=25, deadline=None)
+    @given(n=st.integers(1, 400), L=st.integers(2, 12),
+           seed=st.integers(0, 2**31 - 1), fmax=st.integers(1, 1 << 14))
+    def test_range_coder_property_adaptive(n, L, seed, fmax):
+        """Roundtrip with per-position ADAPTIVE frequency tables (the codec's
+        real operating mode: freqs depend on already-decoded symbols), random
+        alphabet sizes and aggressively skewed counts incl. freq=1 tails."""
+        rng = np.random.default_rng(seed)
+        tables = rng.integers(1, fmax + 1, size=(n, L)).astype(np.int64)
+
+        def freqs_at(i, prev):
+            t = tables[i].copy()
+            if i > 0:            # context-dependence: rotate by prev symbol
+                t = np.roll(t, int(prev))
+            return t
+
+        syms = []
+        for i in range(n):
+            f = freqs_at(i, syms[-1] if syms else 0)
+            p = f / f.sum()
+            syms.append(int(rng.choice(L, p=p)))
+
+        def gen():
+            for i in range(n):
+                yield freqs_at(i, syms[i - 1] if i else 0)
+
+        data = encode_with_freqs(syms, gen())
+        out = decode_with_freqs(
+            data, n, lambda i, dec: freqs_at(i, dec[-1] if dec else 0))
+        assert syms == out

@@ -57,3 +57,39 @@ def test_phase_tables_match_conv_transpose(k, pad, H, W):
                          k - 1 - pad, k - 1 - pad)
     out = _emulate_phase_gather(x, w1, plans, HO, WO, k, k)
     torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-4)
+
+
+# property-based widening: random rectangular kernels / pads / sizes
+try:
+    from hypothesis import given, settings, strategies as st
+    HAVE_HYP = True
+except ImportError:  # pragma: no cover
+    HAVE_HYP = False
+
+
+if HAVE_HYP:
+    @settings(max_examples=30, deadline=None)
+    @given(kh=st.integers(2, 5), kw=st.integers(2, 5),
+           H=st.integers(3, 9), W=st.integers(3, 9),
+           ci=st.integers(1, 4), co=st.integers(1, 5),
+           ph_frac=st.floats(0, 1), pw_frac=st.floats(0, 1),
+           seed=st.integers(0, 2**31 - 1))
+    def test_phase_tables_property(kh, kw, H, W, ci, co, ph_frac, pw_frac,
+                                   seed):
+        """Any (kh,kw,pad_h,pad_w,H,W,Ci,Co) conv-transpose stride-2 geometry
+        must be exactly reproduced by the phase tables (0 <= pad <= k-1, the
+        range where the virtual pads are non-negative — every config shape
+        is in it)."""
+        pad_h = int(ph_frac * (kh - 1))
+        pad_w = int(pw_frac * (kw - 1))
+        torch.manual_seed(seed)
+        x = torch.randn(1, ci, H, W, dtype=torch.float64)
+        w = torch.randn(ci, co, kh, kw, dtype=torch.float64) * 0.3
+        ref = F.conv_transpose2d(x, w, stride=2, padding=(pad_h, pad_w),
+                                 output_padding=1)
+        HO, WO = ref.shape[2], ref.shape[3]
+        w1 = w.flip(2, 3).permute(1, 0, 2, 3).reshape(co, ci * kh * kw)
+        plans = _phase_plans(torch.device("cpu"), HO, WO, kh, kw, ci,
+                             kh - 1 - pad_h, kw - 1 - pad_w)
+        out = _emulate_phase_gather(x, w1, plans, HO, WO, kh, kw)
+        torch.testing.assert_close(out, ref, rtol=1e-9, atol=1e-9)
