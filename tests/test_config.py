@@ -54,3 +54,33 @@ def test_clone_overrides():
 def test_rejects_calls():
     with pytest.raises(ConfigError):
         parse_string("a = __import__('os')\n")
+
+
+try:
+    from hypothesis import given, settings, strategies as st
+    HAVE_HYP = True
+except ImportError:  # pragma: no cover
+    HAVE_HYP = False
+
+
+if HAVE_HYP:
+    _keys = st.from_regex(r"[a-z][a-z0-9_]{0,15}", fullmatch=True)
+    _scalars = st.one_of(
+        st.integers(-10**6, 10**6),
+        st.floats(allow_nan=False, allow_infinity=False, width=32),
+        st.booleans(), st.none(),
+        st.text(st.characters(whitelist_categories=("Ll", "Nd")),
+                min_size=1, max_size=12),
+    )
+    _values = st.one_of(_scalars,
+                        st.tuples(_scalars, _scalars),
+                        st.lists(_scalars, min_size=1, max_size=4))
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.dictionaries(_keys, _values, min_size=1, max_size=12))
+    def test_dump_reparse_value_roundtrip(d):
+        """The sidecar dump (configs_<name>.txt contract) must reparse to the
+        exact same values for every supported value type."""
+        cfg = parse_string("\n".join(f"{k} = {v!r}" for k, v in d.items()))
+        cfg2 = parse_string(str(cfg))
+        assert cfg2.as_dict() == cfg.as_dict() == d
