@@ -125,3 +125,42 @@ def test_nan_guard_blackbox(small_ae_config, pc_config, tmp_path):
     # model + optimizer state present for offline replay
     assert "encoder.h1.conv.weight" in box["model_state"]
     assert len(box["optim_state"]) == 2
+
+
+def test_replay_blackbox_tool(small_ae_config, pc_config, tmp_path, capsys):
+    """tools/replay_blackbox.py localizes the failure from a black box: the
+    poisoned encoder weight must show up as the first non-finite forward."""
+    torch.manual_seed(0)
+    m = DSIN(small_ae_config, pc_config)
+    tr = Trainer(m, small_ae_config, pc_config, num_training_imgs=10,
+                 fused_adam=True, nan_guard=True,
+                 blackbox_dir=str(tmp_path / "bb"))
+    gen = SyntheticStereo(64, 96, seed=3)
+    with torch.no_grad():
+        m.encoder.h1.conv.weight[0, 0, 0, 0] = float("nan")
+    x, y = gen.next_batch()
+    with pytest.warns(UserWarning), pytest.raises(RuntimeError):
+        tr.train_step(x, y)
+    import os
+    box = os.path.join(tmp_path / "bb", os.listdir(tmp_path / "bb")[0])
+
+    # configs via the sidecar dump round-trip (the tool's real input format)
+    aef, pcf = str(tmp_path / "ae"), str(tmp_path / "pc")
+    with open(aef, "w") as f:
+        f.write(str(small_ae_config))
+    with open(pcf, "w") as f:
+        f.write(str(pc_config))
+
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "replay_blackbox",
+        os.path.join(os.path.dirname(__file__), "..", "tools",
+                     "replay_blackbox.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.main([box, "-ae_config", aef, "-pc_config", pcf, "--device", "cpu"])
+    out = capsys.readouterr().out
+    assert "corrupt buffers at dump time" in out
+    assert "encoder.h1.conv.weight" in out       # corruption map names it
+    assert "first non-finite FORWARD output" in out
+    assert "encoder.h1" in out.split("FORWARD output:")[1]
