@@ -156,7 +156,10 @@ def encode_symbols(pc: ProbClass, centers: torch.Tensor,
     pred = PredictionNetwork(pc, centers)
     dev = centers.device
     # ground-truth value buffer: causal reads see the same values the
-    # decoder will have reconstructed; non-causal taps are exact zeros.
+    # decoder will have reconstructed; non-causal positions hold ground
+    # truth here vs the pad value on the decoder side, but every non-causal
+    # tap is multiplied by an exactly-0.0 masked weight, so the per-wave
+    # network outputs are bit-identical anyway (module docstring).
     q_pad = torch.full((C + Dc - 1, H + Hc - 1, W + Wc - 1),
                        float(_pad_value(pc, centers)),
                        dtype=torch.float32, device=dev)
@@ -176,9 +179,9 @@ def decode_symbols(pc: ProbClass, centers: torch.Tensor, data: bytes,
                    shape: Tuple[int, int, int],
                    device: Optional[torch.device] = None) -> torch.Tensor:
     """Wavefront autoregressive decode of a (C, H, W) symbol volume. Exact
-    inverse of encode_symbols(exact=True): one batched network call per
-    skewed wave (O(25C + 5H + W) calls), then the wave's symbols are
-    range-decoded sequentially against the batch's frequency tables."""
+    inverse of encode_symbols: one batched network call per skewed wave
+    (O(25C + 5H + W) calls), then the wave's symbols are range-decoded
+    sequentially against the batch's frequency tables."""
     C, H, W = shape
     device = device or centers.device
     pred = PredictionNetwork(pc, centers)
