@@ -307,6 +307,19 @@ class Trainer:
         per-buffer corruption map locating the first non-finite storage."""
         os.makedirs(self.blackbox_dir, exist_ok=True)
         step = self.global_step
+        # snapshot state BEFORE the diagnostic re-run: train_losses runs in
+        # train mode and would advance the BN running stats, polluting the
+        # at-failure state the box is supposed to preserve
+        model_state = {k: v.cpu().clone() for k, v in
+                       self.model.state_dict().items()}
+        optim_state = [opt.state_dict() for opt in self.optimizers]
+        param_map = {n: self._buf_stats(p)
+                     for n, p in self.model.named_parameters()}
+        opt_map = {}
+        for tag, opt in (("ae", self.opt_ae), ("pc", self.opt_pc)):
+            for buf in ("flat_p", "flat_g", "exp_avg", "exp_avg_sq"):
+                opt_map[f"{tag}.{buf}"] = self._buf_stats(
+                    getattr(opt, buf, None))
         components = {}
         try:
             with self._autocast():
@@ -318,13 +331,6 @@ class Trainer:
                     components[k] = float(v.detach())
         except Exception as e:  # the re-run itself may blow up — still dump
             components["rerun_error"] = repr(e)
-        param_map = {n: self._buf_stats(p)
-                     for n, p in self.model.named_parameters()}
-        opt_map = {}
-        for tag, opt in (("ae", self.opt_ae), ("pc", self.opt_pc)):
-            for buf in ("flat_p", "flat_g", "exp_avg", "exp_avg_sq"):
-                opt_map[f"{tag}.{buf}"] = self._buf_stats(
-                    getattr(opt, buf, None))
         path = os.path.join(self.blackbox_dir, f"step_{step}.pt")
         torch.save({
             "global_step": step,
@@ -332,9 +338,8 @@ class Trainer:
             "components": components,
             "param_stats": param_map,
             "optimizer_stats": opt_map,
-            "model_state": {k: v.cpu() for k, v in
-                            self.model.state_dict().items()},
-            "optim_state": [opt.state_dict() for opt in self.optimizers],
+            "model_state": model_state,
+            "optim_state": optim_state,
         }, path)
         corrupt = {k: v for k, v in {**param_map, **opt_map}.items()
                    if v.get("nonfinite")}
